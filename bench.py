@@ -1,0 +1,302 @@
+#!/usr/bin/env python3
+"""bench.py — BASELINE.json's headline metric on the MI355X hot path.
+
+Metric: "GiB/s erasure encode+bitrot (EC8+4, 1 MiB blocks) per GPU and whole
+node" — one *step* = one fused erasure-encode + per-shard HighwayHash-256
+pass over a batch of independent 1 MiB blocks already resident in HBM
+(BASELINE.json configs[1]: EC8+4, HH256S, batch 1024, 1 GPU).  GiB/s is
+counted on input bytes.
+
+Multi-GPU (--gpus N, launched by the driver via torch.distributed.run): the
+path shards as independent objects — each rank encodes its own batch, no
+data-path collective, barrier + max-over-ranks timing, value
+is the whole-job aggregate.  scaling="weak".
+
+cpu_baseline: the CPU oracle (kind "port" — the reference Go/AVX2 path
+cannot run here, no Go toolchain) timed on host cores over
+a bounded sample.
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--op encode|decode]
+"""
+import argparse
+import ctypes
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+SEED = 0x6D696E696F  # "minio"; synthetic xoshiro256** input (stated in JSON)
+
+# workload = BASELINE.json configs[1] (the metric's quoted configuration)
+WORKLOADS = {
+    "encode": dict(name="EC8+4/1MiB/HH256S/batch1024", d=8, p=4,
+                   bs=1 << 20, n=1024, algo=3, op="encode"),
+    # BASELINE.json configs[3]
+    "decode": dict(name="EC8+4/1MiB/decode-3-erased/batch4096", d=8, p=4,
+                   bs=1 << 20, n=4096, algo=3, op="decode", n_erased=3),
+    # BASELINE.json configs[2]
+    "encode12": dict(name="EC12+4/1MiB/SHA256/batch4096", d=12, p=4,
+                     bs=1 << 20, n=4096, algo=1, op="encode"),
+    # BASELINE.json configs[4] per-GPU slice
+    "encode16": dict(name="EC16+4/4MiB/HH256S/batch512", d=16, p=4,
+                     bs=4 << 20, n=512, algo=3, op="encode"),
+}
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(msg, file=sys.stderr, flush=True)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--op", default="encode", choices=list(WORKLOADS))
+    ap.add_argument("--batch", type=int, default=0, help="override batch/GPU")
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    import minio_amd
+    import oracle
+
+    wl = dict(WORKLOADS[args.op])
+    if args.batch:
+        wl["n"] = args.batch
+    d, p, bs, n, algo = wl["d"], wl["p"], wl["bs"], wl["n"], wl["algo"]
+    total = d + p
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        tdist.init_process_group(backend="gloo")
+        dist = tdist
+
+    if minio_amd.device_count() == 0:
+        raise RuntimeError("bench requires an MI355X GPU")
+
+    S = minio_amd.shard_size(bs, d)
+    stride = minio_amd._lib.mec_shard_stride(bs, d)
+    hsz = 32
+    lib = minio_amd._lib
+
+    e = minio_amd.Erasure(d, p, bs, device=local_rank)
+    ctx = e._ctx
+
+    # ---- stage synthetic inputs into HBM (outside the timed region) ----
+    log(f"[bench] staging {n}x{bs} B on rank {rank} (seed {SEED:#x})")
+    data_bytes = n * d * stride
+    par_bytes = n * p * stride
+    sum_bytes = n * total * hsz
+    vp = ctypes.c_void_p
+
+    def check(st):
+        if st != 0:
+            raise RuntimeError(f"mec status {st}: {lib.mec_last_error()}")
+
+    dev_data, dev_par, dev_sum = vp(), vp(), vp()
+    check(lib.mec_dev_alloc(ctx, data_bytes, ctypes.byref(dev_data)))
+    check(lib.mec_dev_alloc(ctx, par_bytes, ctypes.byref(dev_par)))
+    check(lib.mec_dev_alloc(ctx, sum_bytes, ctypes.byref(dev_sum)))
+
+    # host-generate, scatter into padded strided layout, one H2D copy
+    chunk_blocks = max(1, (256 << 20) // (d * stride))
+    host = ctypes.create_string_buffer(chunk_blocks * d * stride)
+    off = 0
+    bseed = SEED + rank * 1000003
+    blk = bs
+    for b0 in range(0, n, chunk_blocks):
+        nb = min(chunk_blocks, n - b0)
+        for b in range(nb):
+            raw = oracle.fill_random(blk, bseed + b0 + b)
+            for k in range(d):
+                have = max(0, min(S, blk - k * S))
+                dst_off = (b * d + k) * stride
+                host[dst_off:dst_off + have] = raw[k * S:k * S + have]
+                if have < S:
+                    host[dst_off + have:dst_off + S] = b"\0" * (S - have)
+        check(lib.mec_memcpy_h2d(
+            ctx, vp(dev_data.value + off), host, nb * d * stride))
+        off += nb * d * stride
+
+    is_decode = wl["op"] == "decode"
+    present = None
+    if is_decode:
+        # build encoded shard rows in place: data rows + parity rows in one
+        # n x total x stride buffer; erase the first n_erased rows
+        n_er = wl["n_erased"]
+        shards_bytes = n * total * stride
+        dev_shards = vp()
+        check(lib.mec_dev_alloc(ctx, shards_bytes, ctypes.byref(dev_shards)))
+        # encode once to produce parity, then interleave rows as d+p per
+        # item (setup cost, untimed; bounced through host)
+        check(lib.mec_encode_batch_dev(ctx, n, dev_data, bs, dev_par, 0, None))
+        bounce = ctypes.create_string_buffer(d * stride)
+        for b in range(n):
+            check(lib.mec_memcpy_d2h(ctx, bounce,
+                                     vp(dev_data.value + b * d * stride),
+                                     d * stride))
+            check(lib.mec_memcpy_h2d(
+                ctx, vp(dev_shards.value + b * total * stride), bounce,
+                d * stride))
+        bounce2 = ctypes.create_string_buffer(p * stride)
+        for b in range(n):
+            check(lib.mec_memcpy_d2h(ctx, bounce2,
+                                     vp(dev_par.value + b * p * stride),
+                                     p * stride))
+            check(lib.mec_memcpy_h2d(
+                ctx, vp(dev_shards.value + b * total * stride + d * stride),
+                bounce2, p * stride))
+        present = bytes([0] * n_er + [1] * (total - n_er))
+        # zero the erased rows
+        for b in range(n):
+            check(lib.mec_memset_dev(
+                ctx, vp(dev_shards.value + b * total * stride), 0,
+                n_er * stride))
+
+    def step():
+        if is_decode:
+            check(lib.mec_reconstruct_batch_dev_async(
+                ctx, n, dev_shards, present, S, 1))
+        else:
+            check(lib.mec_encode_batch_dev_async(
+                ctx, n, dev_data, bs, dev_par, algo, dev_sum))
+
+    # ---- warmup ----
+    for _ in range(args.warmup):
+        step()
+    check(lib.mec_stream_sync(ctx))
+
+    # ---- timed region: barrier + sync both sides, max over ranks ----
+    if dist:
+        dist.barrier()
+    check(lib.mec_stream_sync(ctx))
+    check(lib.mec_timer_start(ctx))
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    ms = ctypes.c_float()
+    check(lib.mec_timer_stop(ctx, ctypes.byref(ms)))
+    check(lib.mec_stream_sync(ctx))
+    t1 = time.perf_counter()
+    if dist:
+        dist.barrier()
+    wall = t1 - t0
+    gpu_ms = float(ms.value)
+
+    # max over ranks
+    max_wall = wall
+    if dist:
+        import torch
+        t = torch.tensor([wall])
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        max_wall = float(t.item())
+
+    input_bytes_per_step = n * bs * world
+    gib = input_bytes_per_step * args.steps / max_wall / (1 << 30)
+
+    # ---- roofline: dominant-kernel split via HIP events (rank 0, N==1) ----
+    roofline = None
+    if rank == 0:
+        # encode-only leg (GF kernel alone) vs fused to split GF vs hash
+        k = max(4, args.steps // 2)
+        def timed(fn):
+            fn()  # warm
+            check(lib.mec_stream_sync(ctx))
+            check(lib.mec_timer_start(ctx))
+            for _ in range(k):
+                fn()
+            m = ctypes.c_float()
+            check(lib.mec_timer_stop(ctx, ctypes.byref(m)))
+            return float(m.value) / k / 1e3  # s per launch-group
+
+        if is_decode:
+            t_gf = timed(lambda: check(lib.mec_reconstruct_batch_dev_async(
+                ctx, n, dev_shards, present, S, 1)))
+            # algorithmic bytes: read d surviving shards, write n_erased
+            alg_bytes = n * (d * S + wl["n_erased"] * S)
+            dom, t_dom, dom_bytes = "gf_matmul(reconstruct)", t_gf, alg_bytes
+        else:
+            t_gf = timed(lambda: check(lib.mec_encode_batch_dev_async(
+                ctx, n, dev_data, bs, dev_par, algo, None)))
+            t_fused = timed(lambda: check(lib.mec_encode_batch_dev_async(
+                ctx, n, dev_data, bs, dev_par, algo, dev_sum)))
+            t_hash = max(t_fused - t_gf, 1e-9)
+            # per-launch algorithmic bytes:
+            gf_bytes = n * (bs + p * S)          # read data, write parity
+            hash_bytes = n * (total * S)         # read every shard once
+            if t_gf >= t_hash:
+                dom, t_dom, dom_bytes = "gf_matmul(encode)", t_gf, gf_bytes
+            else:
+                dom, t_dom, dom_bytes = "bitrot_hash", t_hash, hash_bytes
+        peak = 8.0e12  # MI355X HBM3E spec peak B/s (MI355X_MICROARCH.md)
+        achieved = dom_bytes / t_dom
+        roofline = {
+            "bound": "hbm", "kernel": dom,
+            "achieved": round(achieved / 1e9, 1), "peak": peak / 1e9,
+            "unit": "GB/s", "frac": round(achieved / peak, 4),
+            "traffic": None,
+        }
+
+    # ---- CPU baseline: oracle timed on host cores (bounded sample) ----
+    cpu_baseline = None
+    if rank == 0 and not args.no_cpu_baseline:
+        lib_o = oracle._lib
+        cores = os.cpu_count() or 1
+        sample_n = 128  # blocks; ~10-30 s of CPU work at 1 MiB blocks
+        log(f"[bench] cpu baseline: oracle, {sample_n} blocks, {cores} threads")
+        if is_decode:
+            el = lib_o.mo_cpu_reconstruct_bench(d, p, bs, sample_n,
+                                                wl["n_erased"], cores, SEED)
+        else:
+            el = lib_o.mo_cpu_encode_bench(d, p, bs, sample_n, algo, cores,
+                                           SEED)
+        if el > 0:
+            cpu_baseline = {
+                "value": round(sample_n * bs / el / (1 << 30), 3),
+                "unit": "GiB/s input", "cores": cores, "kind": "port",
+                "sample": f"{sample_n} x {bs} B blocks "
+                          f"({'reconstruct' if is_decode else 'encode+bitrot'})",
+            }
+
+    if rank == 0:
+        out = {
+            "metric": "GiB/s erasure encode+bitrot (EC8+4, 1 MiB blocks) per GPU and whole node"
+                      if wl["op"] == "encode" and d == 8 else
+                      f"GiB/s erasure {wl['op']} ({wl['name']})",
+            "value": round(gib, 2),
+            "unit": "GiB/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(max_wall * 1e3 / args.steps, 3),
+            "gpu_ms_per_step_rank0": round(gpu_ms / args.steps, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # no published reference number
+            "dtype": "u8",
+            "data": "synthetic",
+            "config": {
+                "workload": wl["name"],
+                "d": d, "p": p, "block_bytes": bs, "batch_per_gpu": n,
+                "bitrot": {1: "sha256", 3: "highwayhash256S"}.get(algo),
+                "seed": hex(SEED), "prng": "xoshiro256**",
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(out), flush=True)
+
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

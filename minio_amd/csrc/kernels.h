@@ -1,0 +1,47 @@
+/* Kernel argument structs shared between ec_abi.cpp (host) and kernels.hip.
+ * Internal to the library — the public boundary is include/minio_ec.h. */
+#ifndef MEC_KERNELS_H
+#define MEC_KERNELS_H
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+/* Kernel-side geometry caps: MinIO erasure sets are 2..16 drives
+ * (docs/distributed/DESIGN.md:44-52); the self-test sweeps d<=14, p<=8
+ * (cmd/erasure-coding.go:152-156).  We allow headroom. */
+#define MEC_KMAX_D 32
+#define MEC_KMAX_E 8  /* output rows per launch; larger p uses multiple */
+#define MEC_KMAX_TOTAL 40
+
+struct GfMatmulArgs {
+    const uint8_t *src;      /* batch base of source rows */
+    uint8_t *dst;            /* batch base of destination rows */
+    int64_t src_item_stride; /* bytes between batch items in src */
+    int64_t dst_item_stride;
+    int64_t row_stride;      /* bytes between shard rows (64-aligned) */
+    int64_t shard_len;       /* valid bytes per shard */
+    int d;                   /* number of source rows */
+    uint8_t src_rows[MEC_KMAX_D];
+    uint8_t dst_rows[MEC_KMAX_E];
+    uint8_t mat[MEC_KMAX_E * MEC_KMAX_D]; /* row t: coefficients over src */
+};
+
+struct HashArgs {
+    const uint8_t *data;   /* data-shard rows, n*d*row_stride */
+    const uint8_t *parity; /* parity rows, n*p*row_stride; NULL => simple
+                              strided layout: chain i at data+i*row_stride */
+    uint8_t *sums;         /* n_chains * digest_size */
+    int64_t row_stride;
+    int64_t msg_len;       /* bytes hashed per chain (the padded shard) */
+    int64_t n_chains;      /* n * (d+p), or n for simple layout */
+    int d, p;
+    uint64_t key[4];       /* HighwayHash key (little-endian words) */
+};
+
+extern "C" {
+hipError_t mec_launch_gf_matmul(const GfMatmulArgs *args, int n_dst, int n,
+                                hipStream_t stream);
+hipError_t mec_launch_hash(int algo, const HashArgs *args, hipStream_t stream);
+}
+
+#endif

@@ -1,0 +1,495 @@
+/* MI355X (gfx950, CDNA4) kernels for the MinIO erasure+bitrot hot path.
+ *
+ * Replaces the compute of:
+ *  - reedsolomon.Encoder.Encode / Reconstruct (GF(2^8) shard arithmetic;
+ *    reference call sites cmd/erasure-coding.go:85,106,112) — gf_matmul
+ *    kernels: byte-wise GF(2^8) constant-multiply as an in-register SWAR
+ *    xtime ladder (poly 0x11D) over 16-byte lanes, wave-uniform coefficient
+ *    bits steering scalar branches.  No MFMA: this is byte/integer work,
+ *    HBM-bound by design.
+ *  - streamingBitrotWriter's per-shard hash (cmd/bitrot-streaming.go:57-59)
+ *    and BitrotAlgorithm.New digests (cmd/bitrot.go:47-64) — one hash chain
+ *    per lane (HighwayHash-256 with the magic key, SHA-256, BLAKE2b-512);
+ *    hashing is sequential per shard, parallelism comes from shards x
+ *    blocks (SURVEY.md §7 hard part (b)).
+ *
+ * Round-1 structure: encode and hash run as back-to-back kernels on one
+ * stream (data re-read by the hash kernel rides L2 for tile-sized batches).
+ * The single-pass fused kernel is the planned round-2 optimization; the
+ * C-ABI signature already treats the pair as one fused operation.
+ */
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+#include "kernels.h"
+
+/* ---- GF(2^8) SWAR helpers --------------------------------------------- */
+
+__device__ __forceinline__ uint32_t gf2x(uint32_t x) {
+    /* multiply 4 packed GF(2^8) bytes by 2 (poly 0x11D) */
+    uint32_t hi = x & 0x80808080u;
+    return ((x & 0x7f7f7f7fu) << 1) ^ ((hi >> 7) * 0x1du);
+}
+
+__device__ __forceinline__ void gf2x4(uint4 &v) {
+    v.x = gf2x(v.x);
+    v.y = gf2x(v.y);
+    v.z = gf2x(v.z);
+    v.w = gf2x(v.w);
+}
+
+__device__ __forceinline__ void xor4(uint4 &a, const uint4 &b) {
+    a.x ^= b.x;
+    a.y ^= b.y;
+    a.z ^= b.z;
+    a.w ^= b.w;
+}
+
+/* ---- generic GF matrix-multiply over shard rows ------------------------
+ *
+ * out[t][j] = sum_k mat[t][k] * src[k][j]  (GF(2^8)), per batch item.
+ * Works for encode (src = data rows, out = parity rows) and reconstruct
+ * (src = surviving rows, out = missing rows) via row index lists.
+ * E = number of output rows (template so accumulators stay in registers).
+ */
+template <int E>
+__global__ void __launch_bounds__(256) gf_matmul_kernel(GfMatmulArgs a) {
+    const int b = blockIdx.y; /* batch item */
+    const int64_t cols = (a.shard_len + 15) >> 4;
+
+    const uint8_t *__restrict__ sbase = a.src + (int64_t)b * a.src_item_stride;
+    uint8_t *__restrict__ obase = a.dst + (int64_t)b * a.dst_item_stride;
+
+    for (int64_t c = blockIdx.x * blockDim.x + threadIdx.x; c < cols;
+         c += (int64_t)gridDim.x * blockDim.x) {
+        const int64_t j = c << 4;
+        uint4 acc[E];
+#pragma unroll
+        for (int i = 0; i < E; i++) acc[i] = uint4{0, 0, 0, 0};
+
+        for (int k = 0; k < a.d; k++) {
+            uint4 pw = *(const uint4 *)(sbase +
+                                        (int64_t)a.src_rows[k] * a.row_stride +
+                                        j);
+            /* wave-uniform coefficient bits -> scalar branches */
+            uint32_t cb[E];
+#pragma unroll
+            for (int i = 0; i < E; i++) cb[i] = a.mat[i * MEC_KMAX_D + k];
+#pragma unroll
+            for (int bit = 0; bit < 8; bit++) {
+                if (bit) gf2x4(pw);
+#pragma unroll
+                for (int i = 0; i < E; i++)
+                    if (cb[i] & (1u << bit)) xor4(acc[i], pw);
+            }
+        }
+#pragma unroll
+        for (int i = 0; i < E; i++)
+            *(uint4 *)(obase + (int64_t)a.dst_rows[i] * a.row_stride + j) =
+                acc[i];
+    }
+}
+
+/* ---- HighwayHash-256 (one chain per lane) ------------------------------
+ * Portable algorithm as published (minio/highwayhash v1.0.3 semantics,
+ * magic key passed from host; pinned by tests/golden/bitrot_selftest.json).
+ */
+
+struct HHState {
+    uint64_t v0[4], v1[4], mul0[4], mul1[4];
+};
+
+__device__ __forceinline__ void hh_zma(uint64_t v1, uint64_t v0,
+                                       uint64_t &add1, uint64_t &add0) {
+    add0 += (((v0 & 0xff000000ull) | (v1 & 0xff00000000ull)) >> 24) |
+            (((v0 & 0xff0000000000ull) | (v1 & 0xff000000000000ull)) >> 16) |
+            (v0 & 0xff0000ull) | ((v0 & 0xff00ull) << 32) |
+            ((v1 & 0xff00000000000000ull) >> 8) | (v0 << 56);
+    add1 += (((v1 & 0xff000000ull) | (v0 & 0xff00000000ull)) >> 24) |
+            (v1 & 0xff0000ull) | ((v1 & 0xff0000000000ull) >> 16) |
+            ((v1 & 0xff00ull) << 24) | ((v0 & 0xff000000000000ull) >> 8) |
+            ((v1 & 0xffull) << 48) | (v0 & 0xff00000000000000ull);
+}
+
+__device__ __forceinline__ void hh_update(HHState &s, const uint64_t lanes[4]) {
+#pragma unroll
+    for (int i = 0; i < 4; i++) {
+        s.v1[i] += s.mul0[i] + lanes[i];
+        s.mul0[i] ^= (s.v1[i] & 0xffffffffull) * (s.v0[i] >> 32);
+        s.v0[i] += s.mul1[i];
+        s.mul1[i] ^= (s.v0[i] & 0xffffffffull) * (s.v1[i] >> 32);
+    }
+    hh_zma(s.v1[1], s.v1[0], s.v0[1], s.v0[0]);
+    hh_zma(s.v1[3], s.v1[2], s.v0[3], s.v0[2]);
+    hh_zma(s.v0[1], s.v0[0], s.v1[1], s.v1[0]);
+    hh_zma(s.v0[3], s.v0[2], s.v1[3], s.v1[2]);
+}
+
+__device__ __forceinline__ void hh_reset(HHState &s, const uint64_t key[4]) {
+    const uint64_t init0[4] = {0xdbe6d5d5fe4cce2full, 0xa4093822299f31d0ull,
+                               0x13198a2e03707344ull, 0x243f6a8885a308d3ull};
+    const uint64_t init1[4] = {0x3bd39e10cb0ef593ull, 0xc0acf169b5f18a8cull,
+                               0xbe5466cf34e90c6cull, 0x452821e638d01377ull};
+#pragma unroll
+    for (int i = 0; i < 4; i++) {
+        s.mul0[i] = init0[i];
+        s.mul1[i] = init1[i];
+        s.v0[i] = init0[i] ^ key[i];
+        s.v1[i] = init1[i] ^ ((key[i] >> 32) | (key[i] << 32));
+    }
+}
+
+__device__ __forceinline__ void hh_finalize256(HHState &s, uint64_t out[4]) {
+#pragma unroll 1
+    for (int r = 0; r < 10; r++) {
+        uint64_t perm[4];
+        perm[0] = (s.v0[2] >> 32) | (s.v0[2] << 32);
+        perm[1] = (s.v0[3] >> 32) | (s.v0[3] << 32);
+        perm[2] = (s.v0[0] >> 32) | (s.v0[0] << 32);
+        perm[3] = (s.v0[1] >> 32) | (s.v0[1] << 32);
+        hh_update(s, perm);
+    }
+    uint64_t a2, a3;
+    a3 = (s.v1[1] + s.mul1[1]) & 0x3fffffffffffffffull;
+    a2 = s.v1[0] + s.mul1[0];
+    out[1] = (s.v0[1] + s.mul0[1]) ^ ((a3 << 1) | (a2 >> 63)) ^
+             ((a3 << 2) | (a2 >> 62));
+    out[0] = (s.v0[0] + s.mul0[0]) ^ (a2 << 1) ^ (a2 << 2);
+    a3 = (s.v1[3] + s.mul1[3]) & 0x3fffffffffffffffull;
+    a2 = s.v1[2] + s.mul1[2];
+    out[3] = (s.v0[3] + s.mul0[3]) ^ ((a3 << 1) | (a2 >> 63)) ^
+             ((a3 << 2) | (a2 >> 62));
+    out[2] = (s.v0[2] + s.mul0[2]) ^ (a2 << 1) ^ (a2 << 2);
+}
+
+/* Map chain index -> shard pointer for the fused encode layout:
+ * chains are (block b, shard s); s < d lives in `data`, else in `parity`. */
+__device__ __forceinline__ const uint8_t *chain_ptr(const HashArgs &a,
+                                                    int64_t chain) {
+    if (a.parity == nullptr) return a.data + chain * a.row_stride;
+    const int total = a.d + a.p;
+    const int64_t b = chain / total;
+    const int s = (int)(chain % total);
+    if (s < a.d)
+        return a.data + (b * a.d + s) * a.row_stride;
+    return a.parity + (b * a.p + (s - a.d)) * a.row_stride;
+}
+
+__global__ void __launch_bounds__(256) hh256_batch_kernel(HashArgs a) {
+    const int64_t chain = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (chain >= a.n_chains) return;
+    const uint8_t *msg = chain_ptr(a, chain);
+
+    HHState s;
+    hh_reset(s, a.key);
+    int64_t len = a.msg_len;
+    /* body: full 32-byte packets via two 16-byte loads */
+    while (len >= 32) {
+        uint4 lo = *(const uint4 *)msg;
+        uint4 hi = *(const uint4 *)(msg + 16);
+        uint64_t lanes[4] = {
+            (uint64_t)lo.x | ((uint64_t)lo.y << 32),
+            (uint64_t)lo.z | ((uint64_t)lo.w << 32),
+            (uint64_t)hi.x | ((uint64_t)hi.y << 32),
+            (uint64_t)hi.z | ((uint64_t)hi.w << 32)};
+        hh_update(s, lanes);
+        msg += 32;
+        len -= 32;
+    }
+    if (len > 0) {
+        /* UpdateRemainder, exactly the published portable semantics */
+        const int mod32 = (int)len;
+        const int mod4 = mod32 & 3;
+#pragma unroll
+        for (int i = 0; i < 4; i++)
+            s.v0[i] += ((uint64_t)mod32 << 32) + (uint64_t)mod32;
+#pragma unroll
+        for (int i = 0; i < 4; i++) {
+            uint32_t h0 = (uint32_t)s.v1[i];
+            uint32_t h1 = (uint32_t)(s.v1[i] >> 32);
+            s.v1[i] = (uint32_t)((h0 << mod32) | (h0 >> (32 - mod32)));
+            s.v1[i] |= (uint64_t)((h1 << mod32) | (h1 >> (32 - mod32))) << 32;
+        }
+        uint8_t packet[32];
+#pragma unroll
+        for (int i = 0; i < 32; i++) packet[i] = 0;
+        for (int i = 0; i < (mod32 & ~3); i++) packet[i] = msg[i];
+        const uint8_t *rem = msg + (mod32 & ~3);
+        if (mod32 & 16) {
+            for (int i = 0; i < 4; i++) packet[28 + i] = rem[i + mod4 - 4];
+        } else if (mod4) {
+            packet[16] = rem[0];
+            packet[17] = rem[mod4 >> 1];
+            packet[18] = rem[mod4 - 1];
+        }
+        uint64_t lanes[4];
+#pragma unroll
+        for (int i = 0; i < 4; i++) {
+            uint64_t v = 0;
+            for (int bt = 7; bt >= 0; bt--) v = (v << 8) | packet[8 * i + bt];
+            lanes[i] = v;
+        }
+        hh_update(s, lanes);
+    }
+    uint64_t out[4];
+    hh_finalize256(s, out);
+    *(uint64_t *)(a.sums + chain * 32 + 0) = out[0];
+    *(uint64_t *)(a.sums + chain * 32 + 8) = out[1];
+    *(uint64_t *)(a.sums + chain * 32 + 16) = out[2];
+    *(uint64_t *)(a.sums + chain * 32 + 24) = out[3];
+}
+
+/* ---- SHA-256 (one chain per lane), FIPS 180-4 -------------------------- */
+
+__constant__ uint32_t SHA_K[64] = {
+    0x428a2f98, 0x71374491, 0xb5c0fbcf, 0xe9b5dba5, 0x3956c25b, 0x59f111f1,
+    0x923f82a4, 0xab1c5ed5, 0xd807aa98, 0x12835b01, 0x243185be, 0x550c7dc3,
+    0x72be5d74, 0x80deb1fe, 0x9bdc06a7, 0xc19bf174, 0xe49b69c1, 0xefbe4786,
+    0x0fc19dc6, 0x240ca1cc, 0x2de92c6f, 0x4a7484aa, 0x5cb0a9dc, 0x76f988da,
+    0x983e5152, 0xa831c66d, 0xb00327c8, 0xbf597fc7, 0xc6e00bf3, 0xd5a79147,
+    0x06ca6351, 0x14292967, 0x27b70a85, 0x2e1b2138, 0x4d2c6dfc, 0x53380d13,
+    0x650a7354, 0x766a0abb, 0x81c2c92e, 0x92722c85, 0xa2bfe8a1, 0xa81a664b,
+    0xc24b8b70, 0xc76c51a3, 0xd192e819, 0xd6990624, 0xf40e3585, 0x106aa070,
+    0x19a4c116, 0x1e376c08, 0x2748774c, 0x34b0bcb5, 0x391c0cb3, 0x4ed8aa4a,
+    0x5b9cca4f, 0x682e6ff3, 0x748f82ee, 0x78a5636f, 0x84c87814, 0x8cc70208,
+    0x90befffa, 0xa4506ceb, 0xbef9a3f7, 0xc67178f2};
+
+__device__ __forceinline__ uint32_t rotr32(uint32_t x, int n) {
+    return (x >> n) | (x << (32 - n));
+}
+
+__device__ __forceinline__ uint32_t bswap32(uint32_t v) {
+    return __builtin_bswap32(v);
+}
+
+__device__ void sha256_block(uint32_t h[8], const uint32_t w_in[16]) {
+    uint32_t w[16];
+#pragma unroll
+    for (int i = 0; i < 16; i++) w[i] = w_in[i];
+    uint32_t a = h[0], b = h[1], c = h[2], d = h[3], e = h[4], f = h[5],
+             g = h[6], hh = h[7];
+#pragma unroll 4
+    for (int i = 0; i < 64; i++) {
+        uint32_t wi;
+        if (i < 16) {
+            wi = w[i];
+        } else {
+            uint32_t w15 = w[(i - 15) & 15], w2 = w[(i - 2) & 15];
+            uint32_t s0 = rotr32(w15, 7) ^ rotr32(w15, 18) ^ (w15 >> 3);
+            uint32_t s1 = rotr32(w2, 17) ^ rotr32(w2, 19) ^ (w2 >> 10);
+            wi = w[i & 15] + s0 + w[(i - 7) & 15] + s1;
+            w[i & 15] = wi;
+        }
+        uint32_t S1 = rotr32(e, 6) ^ rotr32(e, 11) ^ rotr32(e, 25);
+        uint32_t ch = (e & f) ^ (~e & g);
+        uint32_t t1 = hh + S1 + ch + SHA_K[i] + wi;
+        uint32_t S0 = rotr32(a, 2) ^ rotr32(a, 13) ^ rotr32(a, 22);
+        uint32_t maj = (a & b) ^ (a & c) ^ (b & c);
+        uint32_t t2 = S0 + maj;
+        hh = g; g = f; f = e; e = d + t1;
+        d = c; c = b; b = a; a = t1 + t2;
+    }
+    h[0] += a; h[1] += b; h[2] += c; h[3] += d;
+    h[4] += e; h[5] += f; h[6] += g; h[7] += hh;
+}
+
+__global__ void __launch_bounds__(256) sha256_batch_kernel(HashArgs a) {
+    const int64_t chain = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (chain >= a.n_chains) return;
+    const uint8_t *msg = chain_ptr(a, chain);
+
+    uint32_t h[8] = {0x6a09e667, 0xbb67ae85, 0x3c6ef372, 0xa54ff53a,
+                     0x510e527f, 0x9b05688c, 0x1f83d9ab, 0x5be0cd19};
+    int64_t len = a.msg_len;
+    uint32_t w[16];
+    while (len >= 64) {
+        const uint4 *p = (const uint4 *)msg;
+#pragma unroll
+        for (int q = 0; q < 4; q++) {
+            uint4 v = p[q];
+            w[4 * q + 0] = bswap32(v.x);
+            w[4 * q + 1] = bswap32(v.y);
+            w[4 * q + 2] = bswap32(v.z);
+            w[4 * q + 3] = bswap32(v.w);
+        }
+        sha256_block(h, w);
+        msg += 64;
+        len -= 64;
+    }
+    /* tail: rem bytes + 0x80 pad + 8-byte big-endian bit length */
+    {
+        uint8_t tail[128];
+#pragma unroll
+        for (int i = 0; i < 128; i++) tail[i] = 0;
+        for (int i = 0; i < (int)len; i++) tail[i] = msg[i];
+        tail[(int)len] = 0x80;
+        const int tlen = (len < 56) ? 64 : 128;
+        uint64_t bits = (uint64_t)a.msg_len * 8;
+        for (int i = 0; i < 8; i++)
+            tail[tlen - 1 - i] = (uint8_t)(bits >> (8 * i));
+        for (int blk = 0; blk < tlen; blk += 64) {
+            for (int i = 0; i < 16; i++) {
+                const uint8_t *q = tail + blk + 4 * i;
+                w[i] = ((uint32_t)q[0] << 24) | ((uint32_t)q[1] << 16) |
+                       ((uint32_t)q[2] << 8) | q[3];
+            }
+            sha256_block(h, w);
+        }
+    }
+    uint8_t *out = a.sums + chain * 32;
+    for (int i = 0; i < 8; i++)
+        *(uint32_t *)(out + 4 * i) = bswap32(h[i]);
+}
+
+/* ---- BLAKE2b-512 (one chain per lane), RFC 7693 ------------------------ */
+
+__constant__ uint64_t B2B_IV[8] = {
+    0x6a09e667f3bcc908ull, 0xbb67ae8584caa73bull, 0x3c6ef372fe94f82bull,
+    0xa54ff53a5f1d36f1ull, 0x510e527fade682d1ull, 0x9b05688c2b3e6c1full,
+    0x1f83d9abfb41bd6bull, 0x5be0cd19137e2179ull};
+
+__constant__ uint8_t B2B_SIGMA[12][16] = {
+    {0, 1, 2, 3, 4, 5, 6, 7, 8, 9, 10, 11, 12, 13, 14, 15},
+    {14, 10, 4, 8, 9, 15, 13, 6, 1, 12, 0, 2, 11, 7, 5, 3},
+    {11, 8, 12, 0, 5, 2, 15, 13, 10, 14, 3, 6, 7, 1, 9, 4},
+    {7, 9, 3, 1, 13, 12, 11, 14, 2, 6, 5, 10, 4, 0, 15, 8},
+    {9, 0, 5, 7, 2, 4, 10, 15, 14, 1, 11, 12, 6, 8, 3, 13},
+    {2, 12, 6, 10, 0, 11, 8, 3, 4, 13, 7, 5, 15, 14, 1, 9},
+    {12, 5, 1, 15, 14, 13, 4, 10, 0, 7, 6, 3, 9, 2, 8, 11},
+    {13, 11, 7, 14, 12, 1, 3, 9, 5, 0, 15, 4, 8, 6, 2, 10},
+    {6, 15, 14, 9, 11, 3, 0, 8, 12, 2, 13, 7, 1, 4, 10, 5},
+    {10, 2, 8, 4, 7, 6, 1, 5, 15, 11, 9, 14, 3, 12, 13, 0},
+    {0, 1, 2, 3, 4, 5, 6, 7, 8, 9, 10, 11, 12, 13, 14, 15},
+    {14, 10, 4, 8, 9, 15, 13, 6, 1, 12, 0, 2, 11, 7, 5, 3}};
+
+__device__ __forceinline__ uint64_t rotr64d(uint64_t x, int n) {
+    return (x >> n) | (x << (64 - n));
+}
+
+__device__ void b2b_compress(uint64_t h[8], const uint64_t m[16], uint64_t t,
+                             bool last) {
+    uint64_t v[16];
+#pragma unroll
+    for (int i = 0; i < 8; i++) v[i] = h[i];
+#pragma unroll
+    for (int i = 0; i < 8; i++) v[8 + i] = B2B_IV[i];
+    v[12] ^= t;
+    if (last) v[14] = ~v[14];
+#pragma unroll 1
+    for (int r = 0; r < 12; r++) {
+        const uint8_t *sg = B2B_SIGMA[r];
+#define B2B_G(ai, bi, ci, di, x, y)                                          \
+    {                                                                        \
+        v[ai] = v[ai] + v[bi] + (x);                                         \
+        v[di] = rotr64d(v[di] ^ v[ai], 32);                                  \
+        v[ci] = v[ci] + v[di];                                               \
+        v[bi] = rotr64d(v[bi] ^ v[ci], 24);                                  \
+        v[ai] = v[ai] + v[bi] + (y);                                         \
+        v[di] = rotr64d(v[di] ^ v[ai], 16);                                  \
+        v[ci] = v[ci] + v[di];                                               \
+        v[bi] = rotr64d(v[bi] ^ v[ci], 63);                                  \
+    }
+        B2B_G(0, 4, 8, 12, m[sg[0]], m[sg[1]]);
+        B2B_G(1, 5, 9, 13, m[sg[2]], m[sg[3]]);
+        B2B_G(2, 6, 10, 14, m[sg[4]], m[sg[5]]);
+        B2B_G(3, 7, 11, 15, m[sg[6]], m[sg[7]]);
+        B2B_G(0, 5, 10, 15, m[sg[8]], m[sg[9]]);
+        B2B_G(1, 6, 11, 12, m[sg[10]], m[sg[11]]);
+        B2B_G(2, 7, 8, 13, m[sg[12]], m[sg[13]]);
+        B2B_G(3, 4, 9, 14, m[sg[14]], m[sg[15]]);
+#undef B2B_G
+    }
+#pragma unroll
+    for (int i = 0; i < 8; i++) h[i] ^= v[i] ^ v[8 + i];
+}
+
+__global__ void __launch_bounds__(256) blake2b512_batch_kernel(HashArgs a) {
+    const int64_t chain = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (chain >= a.n_chains) return;
+    const uint8_t *msg = chain_ptr(a, chain);
+
+    uint64_t h[8];
+#pragma unroll
+    for (int i = 0; i < 8; i++) h[i] = B2B_IV[i];
+    h[0] ^= 0x01010000ull ^ 64;
+    int64_t off = 0;
+    uint64_t m[16];
+    while (a.msg_len - off > 128) {
+        const uint4 *p = (const uint4 *)(msg + off);
+#pragma unroll
+        for (int q = 0; q < 8; q++) {
+            uint4 v = p[q];
+            m[2 * q] = (uint64_t)v.x | ((uint64_t)v.y << 32);
+            m[2 * q + 1] = (uint64_t)v.z | ((uint64_t)v.w << 32);
+        }
+        b2b_compress(h, m, (uint64_t)(off + 128), false);
+        off += 128;
+    }
+    {
+        const int rem = (int)(a.msg_len - off);
+        uint8_t tail[128];
+#pragma unroll
+        for (int i = 0; i < 128; i++) tail[i] = 0;
+        for (int i = 0; i < rem; i++) tail[i] = msg[off + i];
+        for (int i = 0; i < 16; i++) {
+            uint64_t v = 0;
+            for (int bt = 7; bt >= 0; bt--) v = (v << 8) | tail[8 * i + bt];
+            m[i] = v;
+        }
+        b2b_compress(h, m, (uint64_t)a.msg_len, true);
+    }
+    uint8_t *out = a.sums + chain * 64;
+    for (int i = 0; i < 8; i++) *(uint64_t *)(out + 8 * i) = h[i];
+}
+
+/* ---- launch wrappers (called from ec_abi.cpp) -------------------------- */
+
+extern "C" {
+
+hipError_t mec_launch_gf_matmul(const GfMatmulArgs *args, int n_dst, int n,
+                                hipStream_t stream) {
+    const int64_t cols = (args->shard_len + 15) >> 4;
+    /* >=2048 workgroups total to fill 256 CUs, but never more than the work */
+    int64_t max_x = (cols + 255) / 256;
+    int64_t want_x = (2048 + n - 1) / n;
+    int64_t blocks_x = want_x < max_x ? want_x : max_x;
+    if (blocks_x < 1) blocks_x = 1;
+    dim3 grid((uint32_t)blocks_x, n);
+    dim3 blk(256);
+#define CASE(E)                                                              \
+    case E:                                                                  \
+        hipLaunchKernelGGL(gf_matmul_kernel<E>, grid, blk, 0, stream,        \
+                           *args);                                           \
+        break;
+    switch (n_dst) {
+        CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
+    default:
+        return hipErrorInvalidValue;
+    }
+#undef CASE
+    return hipGetLastError();
+}
+
+hipError_t mec_launch_hash(int algo, const HashArgs *args,
+                           hipStream_t stream) {
+    const int64_t blocks = (args->n_chains + 255) / 256;
+    dim3 grid((uint32_t)blocks);
+    dim3 blk(256);
+    switch (algo) {
+    case 1: /* SHA256 */
+        hipLaunchKernelGGL(sha256_batch_kernel, grid, blk, 0, stream, *args);
+        break;
+    case 2: /* HighwayHash256 */
+    case 3: /* HighwayHash256S */
+        hipLaunchKernelGGL(hh256_batch_kernel, grid, blk, 0, stream, *args);
+        break;
+    case 4: /* BLAKE2b512 */
+        hipLaunchKernelGGL(blake2b512_batch_kernel, grid, blk, 0, stream,
+                           *args);
+        break;
+    default:
+        return hipErrorInvalidValue;
+    }
+    return hipGetLastError();
+}
+}

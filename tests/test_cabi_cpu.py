@@ -1,0 +1,63 @@
+"""CPU-side checks of the product library: it loads, exports every symbol
+include/minio_ec.h declares, its pure host math mirrors the reference
+formulas, and GPU entry points fail LOUDLY (NoGPUError) with no GPU —
+never a silent fallback."""
+import ctypes
+import os
+import re
+
+import pytest
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+REPO = os.path.dirname(HERE)
+
+import minio_amd
+import oracle
+
+
+def test_exports_match_header():
+    hdr = open(os.path.join(REPO, "include/minio_ec.h")).read()
+    # function names: declarations ending in '(' at top level
+    names = re.findall(r"\b(mec_[a-z0-9_]+)\s*\(", hdr)
+    names = sorted(set(names))
+    assert len(names) > 25
+    lib = ctypes.CDLL(os.path.join(REPO, "minio_amd/libminio_ec_hip.so"))
+    for n in names:
+        assert hasattr(lib, n), f"symbol {n} missing from libminio_ec_hip.so"
+
+
+def test_shard_math_mirrors_reference():
+    # ShardSize / ShardFileSize / ShardFileOffset (cmd/erasure-coding.go:116-141)
+    bs = 1 << 20
+    assert minio_amd.shard_size(bs, 8) == 131072
+    assert minio_amd.shard_size(bs, 12) == 87382
+    assert minio_amd.shard_file_size(bs, 8, 0) == 0
+    assert minio_amd.shard_file_size(bs, 8, -1) == -1
+    total = 2 * bs + 12345
+    assert minio_amd.shard_file_size(bs, 8, total) == 2 * 131072 + oracle.ceil_frac(12345, 8)
+    # bitrotShardFileSize (cmd/bitrot.go:156-161); test shape from
+    # cmd/bitrot_test.go:38 (length 35, shardSize 10 -> 4 hashes)
+    assert minio_amd.bitrot_shard_file_size(35, 10, minio_amd.HIGHWAYHASH256S) == 4 * 32 + 35
+    assert minio_amd.bitrot_shard_file_size(35, 10, minio_amd.SHA256) == 35
+    # ShardFileOffset
+    got = minio_amd.shard_file_offset(bs, 8, 0, total, total)
+    assert got == minio_amd.shard_file_size(bs, 8, total)
+
+
+def test_no_gpu_fails_loudly():
+    if minio_amd.device_count() > 0:
+        pytest.skip("GPU present")
+    with pytest.raises(minio_amd.NoGPUError):
+        minio_amd.Erasure(8, 4, 1 << 20)
+
+
+def test_geometry_validation():
+    if minio_amd.device_count() == 0:
+        # validation precedes GPU checks for clearly-invalid geometries
+        lib = minio_amd._lib
+        ctx = ctypes.c_void_p()
+        assert lib.mec_ctx_create(0, 2, 1 << 20, 0, ctypes.byref(ctx)) == 1
+        assert lib.mec_ctx_create(200, 100, 1 << 20, 0, ctypes.byref(ctx)) == 2
+    else:
+        with pytest.raises(minio_amd.MecError):
+            minio_amd.Erasure(0, 2)

@@ -1,0 +1,101 @@
+"""Size-independent properties of the oracle, mirroring the reference's
+behavioural tests (cmd/erasure_test.go:45-108 round trips,
+cmd/erasure-heal_test.go erase patterns, cmd/bitrot_test.go shapes)."""
+import itertools
+import random
+
+import pytest
+
+import oracle
+
+
+@pytest.mark.parametrize("d,p", [(2, 2), (4, 2), (8, 4), (12, 4), (16, 4)])
+def test_roundtrip_erase_reconstruct(d, p):
+    rng = random.Random(0x6D696E696F + d * 100 + p)
+    data = bytes(rng.getrandbits(8) for _ in range(4096 + 13))
+    rs = oracle.RS(d, p)
+    shards = rs.encode_data(data)
+    # every erase pattern of size p over a few random choices
+    idxs = list(range(d + p))
+    for _ in range(10):
+        erased = rng.sample(idxs, p)
+        damaged = [None if i in erased else s for i, s in enumerate(shards)]
+        rec = rs.reconstruct(damaged, data_only=False)
+        assert rec == shards, f"erased={erased}"
+
+
+def test_split_padding_semantics():
+    # Split: shard k = bytes [k*ceil(n/d), ...), tail zero-padded
+    # (cmd/erasure-coding.go:81 -> reedsolomon Split)
+    rs = oracle.RS(3, 2)
+    data = bytes((i + 1) & 0xFF for i in range(256))  # 256 B, d=3 -> per=86, pad=2
+    shards = rs.encode_data(data)
+    assert len(shards[0]) == 86
+    assert shards[0] == data[:86]
+    assert shards[1] == data[86:172]
+    assert shards[2] == data[172:] + b"\0\0"
+
+
+def test_reconstruct_too_few():
+    rs = oracle.RS(4, 2)
+    shards = rs.encode_data(bytes(64))
+    damaged = [None, None, None] + shards[3:]
+    with pytest.raises(ValueError):
+        rs.reconstruct(damaged)
+
+
+def test_data_only_leaves_parity_missing():
+    rs = oracle.RS(4, 2)
+    shards = rs.encode_data(bytes(range(64)))
+    damaged = [None] + shards[1:5] + [None]
+    rec = rs.reconstruct(damaged, data_only=True)
+    assert rec[0] == shards[0]
+    assert rec[5] is None
+
+
+def test_hh256_ragged_tails_vs_full():
+    # ragged lengths exercise UpdateRemainder; shape check only (the exact
+    # values are pinned transitively -- see oracle.h header note)
+    for n in [0, 1, 3, 4, 15, 16, 17, 31, 32, 33, 63, 64, 100]:
+        msg = bytes((i * 7 + 1) & 0xFF for i in range(n))
+        s1 = oracle.bitrot_sum(oracle.HIGHWAYHASH256S, msg)
+        s2 = oracle.bitrot_sum(oracle.HIGHWAYHASH256S, msg)
+        assert s1 == s2 and len(s1) == 32
+        if n > 0:
+            s3 = oracle.bitrot_sum(oracle.HIGHWAYHASH256S, msg[:-1] + bytes([msg[-1] ^ 1]))
+            assert s3 != s1
+
+
+def test_shard_file_size_math():
+    # cmd/erasure-coding.go:121-132 and cmd/bitrot.go:156-161
+    bs, d = 1 << 20, 8
+    S = oracle.ceil_frac(bs, d)
+    assert S == 131072
+    # 2.5 blocks
+    total = 2 * bs + 12345
+    want = 2 * S + oracle.ceil_frac(12345, d)
+    got = (total // bs) * S + oracle.ceil_frac(total % bs, d)
+    assert got == want
+    assert oracle.bitrot_shard_file_size(35, 10, oracle.HIGHWAYHASH256S) == 4 * 32 + 35
+    assert oracle.bitrot_shard_file_size(35, 10, oracle.SHA256) == 35
+
+
+def test_encode_stream_layout():
+    # [hash||shard]* per drive (cmd/bitrot-streaming.go:57-75)
+    d, p, bs = 4, 2, 256
+    data = oracle.fill_random(1000, 42)
+    streams, _ = oracle.encode_stream(d, p, bs, data, oracle.HIGHWAYHASH256S)
+    S = oracle.ceil_frac(bs, d)  # 64
+    n_blocks = 4  # ceil(1000/256); last block 232 bytes -> S_last = 58
+    rs = oracle.RS(d, p)
+    for i, st in enumerate(streams):
+        off = 0
+        for b in range(n_blocks):
+            blk = data[b * bs:(b + 1) * bs]
+            shards = rs.encode_data(blk)
+            h = st[off:off + 32]
+            sh = st[off + 32:off + 32 + len(shards[i])]
+            assert sh == shards[i]
+            assert h == oracle.bitrot_sum(oracle.HIGHWAYHASH256S, sh)
+            off += 32 + len(shards[i])
+        assert off == len(st)
