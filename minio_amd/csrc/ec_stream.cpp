@@ -196,7 +196,12 @@ mec_status mec_decode_stream(mec_ctx *ctx_, const uint8_t *const *drive_bufs,
     }
 
     int64_t start_block = offset / block_size;
-    int64_t end_block = (offset + length) / block_size;
+    /* the reference iterates to endBlock = (offset+length)/blockSize
+     * inclusive and breaks on a zero-length final block
+     * (cmd/erasure-decode.go:257-280); reads only touch blocks that hold
+     * data */
+    int64_t end_block_raw = (offset + length) / block_size;
+    int64_t end_block = end_block_raw;
     if (end_block >= g.n_blocks) end_block = g.n_blocks - 1;
 
     /* verify-on-read for the streaming format, batched per drive over the
@@ -297,16 +302,15 @@ mec_status mec_decode_stream(mec_ctx *ctx_, const uint8_t *const *drive_bufs,
     /* writeDataBlocks: trim [offset, offset+length) out of the block
      * stream (cmd/erasure-utils.go:42-105 via cmd/erasure-decode.go:262) */
     int64_t written = 0;
-    for (int64_t b = 0; b < nb; b++) {
-        int64_t gb = start_block + b;
+    for (int64_t gb = start_block; gb <= end_block_raw; gb++) {
         int64_t block_off, block_len;
-        if (start_block == end_block) {
+        if (start_block == end_block_raw) {
             block_off = offset % block_size;
             block_len = length;
         } else if (gb == start_block) {
             block_off = offset % block_size;
             block_len = block_size - block_off;
-        } else if (gb == end_block) {
+        } else if (gb == end_block_raw) {
             block_off = 0;
             block_len = (offset + length) % block_size;
         } else {
@@ -314,7 +318,8 @@ mec_status mec_decode_stream(mec_ctx *ctx_, const uint8_t *const *drive_bufs,
             block_len = block_size;
         }
         if (block_len == 0) break;
-        memcpy(dst + written, block_data[(size_t)b].data() + block_off,
+        memcpy(dst + written,
+               block_data[(size_t)(gb - start_block)].data() + block_off,
                (size_t)block_len);
         written += block_len;
     }

@@ -175,46 +175,105 @@ __device__ __forceinline__ const uint8_t *chain_ptr(const HashArgs &a,
     return a.parity + (b * a.p + (s - a.d)) * a.row_stride;
 }
 
+/* 4-lane HighwayHash: one chain per 4 consecutive GPU lanes, lane li owns
+ * HighwayHash lane li of the state.  Zipper merges and the finalization
+ * permute exchange partner words via __shfl_xor inside the quad.  4x the
+ * chain parallelism of a per-lane design and 1/4 the serial work per lane —
+ * hashing is sequential per shard, so chains x lanes is the only
+ * parallelism available (SURVEY.md §7 hard part (b)). */
+
+__device__ __forceinline__ uint64_t zip_even(uint64_t v0, uint64_t v1) {
+    /* add0 expression of ZipperMergeAndAdd(v1=v1[odd], v0=v1[even], ...) */
+    return (((v0 & 0xff000000ull) | (v1 & 0xff00000000ull)) >> 24) |
+           (((v0 & 0xff0000000000ull) | (v1 & 0xff000000000000ull)) >> 16) |
+           (v0 & 0xff0000ull) | ((v0 & 0xff00ull) << 32) |
+           ((v1 & 0xff00000000000000ull) >> 8) | (v0 << 56);
+}
+
+__device__ __forceinline__ uint64_t zip_odd(uint64_t v0, uint64_t v1) {
+    /* add1 expression; v0 = v1[even] (partner), v1 = v1[odd] (own) */
+    return (((v1 & 0xff000000ull) | (v0 & 0xff00000000ull)) >> 24) |
+           (v1 & 0xff0000ull) | ((v1 & 0xff0000000000ull) >> 16) |
+           ((v1 & 0xff00ull) << 24) | ((v0 & 0xff000000000000ull) >> 8) |
+           ((v1 & 0xffull) << 48) | (v0 & 0xff00000000000000ull);
+}
+
+struct HH4 {
+    uint64_t v0, v1, mul0, mul1; /* this GPU lane's HighwayHash lane */
+};
+
+__device__ __forceinline__ uint64_t shfl_x(uint64_t v, int mask) {
+    return __shfl_xor((unsigned long long)v, mask, 64);
+}
+
+__device__ __forceinline__ void hh4_update(HH4 &s, uint64_t lane_word,
+                                           bool odd) {
+    s.v1 += s.mul0 + lane_word;
+    s.mul0 ^= (s.v1 & 0xffffffffull) * (s.v0 >> 32);
+    s.v0 += s.mul1;
+    s.mul1 ^= (s.v0 & 0xffffffffull) * (s.v1 >> 32);
+    uint64_t pv1 = shfl_x(s.v1, 1);
+    s.v0 += odd ? zip_odd(pv1, s.v1) : zip_even(s.v1, pv1);
+    uint64_t pv0 = shfl_x(s.v0, 1);
+    s.v1 += odd ? zip_odd(pv0, s.v0) : zip_even(s.v0, pv0);
+}
+
 __global__ void __launch_bounds__(256) hh256_batch_kernel(HashArgs a) {
-    const int64_t chain = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t chain = tid >> 2;
+    const int li = (int)(tid & 3); /* HighwayHash lane index */
     if (chain >= a.n_chains) return;
     const uint8_t *msg = chain_ptr(a, chain);
+    const bool odd = li & 1;
 
-    HHState s;
-    hh_reset(s, a.key);
+    const uint64_t init0[4] = {0xdbe6d5d5fe4cce2full, 0xa4093822299f31d0ull,
+                               0x13198a2e03707344ull, 0x243f6a8885a308d3ull};
+    const uint64_t init1[4] = {0x3bd39e10cb0ef593ull, 0xc0acf169b5f18a8cull,
+                               0xbe5466cf34e90c6cull, 0x452821e638d01377ull};
+    HH4 s;
+    s.mul0 = init0[li];
+    s.mul1 = init1[li];
+    s.v0 = init0[li] ^ a.key[li];
+    s.v1 = init1[li] ^ ((a.key[li] >> 32) | (a.key[li] << 32));
+
     int64_t len = a.msg_len;
-    /* body: full 32-byte packets via two 16-byte loads */
+    const uint8_t *mp = msg + 8 * li;
+    /* unrolled by 4 packets for memory-level parallelism */
+    while (len >= 128) {
+        uint64_t w0 = *(const uint64_t *)(mp + 0);
+        uint64_t w1 = *(const uint64_t *)(mp + 32);
+        uint64_t w2 = *(const uint64_t *)(mp + 64);
+        uint64_t w3 = *(const uint64_t *)(mp + 96);
+        hh4_update(s, w0, odd);
+        hh4_update(s, w1, odd);
+        hh4_update(s, w2, odd);
+        hh4_update(s, w3, odd);
+        mp += 128;
+        len -= 128;
+    }
     while (len >= 32) {
-        uint4 lo = *(const uint4 *)msg;
-        uint4 hi = *(const uint4 *)(msg + 16);
-        uint64_t lanes[4] = {
-            (uint64_t)lo.x | ((uint64_t)lo.y << 32),
-            (uint64_t)lo.z | ((uint64_t)lo.w << 32),
-            (uint64_t)hi.x | ((uint64_t)hi.y << 32),
-            (uint64_t)hi.z | ((uint64_t)hi.w << 32)};
-        hh_update(s, lanes);
-        msg += 32;
+        hh4_update(s, *(const uint64_t *)mp, odd);
+        mp += 32;
         len -= 32;
     }
     if (len > 0) {
-        /* UpdateRemainder, exactly the published portable semantics */
+        /* UpdateRemainder (published portable semantics); each lane builds
+         * the full 32-byte packet locally — tail-only cost */
+        const uint8_t *tail_msg = msg + (a.msg_len - len);
         const int mod32 = (int)len;
         const int mod4 = mod32 & 3;
-#pragma unroll
-        for (int i = 0; i < 4; i++)
-            s.v0[i] += ((uint64_t)mod32 << 32) + (uint64_t)mod32;
-#pragma unroll
-        for (int i = 0; i < 4; i++) {
-            uint32_t h0 = (uint32_t)s.v1[i];
-            uint32_t h1 = (uint32_t)(s.v1[i] >> 32);
-            s.v1[i] = (uint32_t)((h0 << mod32) | (h0 >> (32 - mod32)));
-            s.v1[i] |= (uint64_t)((h1 << mod32) | (h1 >> (32 - mod32))) << 32;
+        s.v0 += ((uint64_t)mod32 << 32) + (uint64_t)mod32;
+        {
+            uint32_t h0 = (uint32_t)s.v1;
+            uint32_t h1 = (uint32_t)(s.v1 >> 32);
+            s.v1 = (uint32_t)((h0 << mod32) | (h0 >> (32 - mod32)));
+            s.v1 |= (uint64_t)((h1 << mod32) | (h1 >> (32 - mod32))) << 32;
         }
         uint8_t packet[32];
 #pragma unroll
         for (int i = 0; i < 32; i++) packet[i] = 0;
-        for (int i = 0; i < (mod32 & ~3); i++) packet[i] = msg[i];
-        const uint8_t *rem = msg + (mod32 & ~3);
+        for (int i = 0; i < (mod32 & ~3); i++) packet[i] = tail_msg[i];
+        const uint8_t *rem = tail_msg + (mod32 & ~3);
         if (mod32 & 16) {
             for (int i = 0; i < 4; i++) packet[28 + i] = rem[i + mod4 - 4];
         } else if (mod4) {
@@ -222,21 +281,31 @@ __global__ void __launch_bounds__(256) hh256_batch_kernel(HashArgs a) {
             packet[17] = rem[mod4 >> 1];
             packet[18] = rem[mod4 - 1];
         }
-        uint64_t lanes[4];
-#pragma unroll
-        for (int i = 0; i < 4; i++) {
-            uint64_t v = 0;
-            for (int bt = 7; bt >= 0; bt--) v = (v << 8) | packet[8 * i + bt];
-            lanes[i] = v;
-        }
-        hh_update(s, lanes);
+        uint64_t w = 0;
+        for (int bt = 7; bt >= 0; bt--) w = (w << 8) | packet[8 * li + bt];
+        hh4_update(s, w, odd);
     }
-    uint64_t out[4];
-    hh_finalize256(s, out);
-    *(uint64_t *)(a.sums + chain * 32 + 0) = out[0];
-    *(uint64_t *)(a.sums + chain * 32 + 8) = out[1];
-    *(uint64_t *)(a.sums + chain * 32 + 16) = out[2];
-    *(uint64_t *)(a.sums + chain * 32 + 24) = out[3];
+    /* finalization: 10 permute-update rounds; permuted lane li reads
+     * rot32(v0[li ^ 2]) */
+#pragma unroll 1
+    for (int r = 0; r < 10; r++) {
+        uint64_t pv = shfl_x(s.v0, 2);
+        hh4_update(s, (pv >> 32) | (pv << 32), odd);
+    }
+    /* modular reduction: even lanes all-local; odd lanes need partner's
+     * (v1+mul1) */
+    uint64_t sv = s.v1 + s.mul1;
+    uint64_t sv0 = s.v0 + s.mul0;
+    uint64_t sv_part = shfl_x(sv, 1);
+    uint64_t out;
+    if (!odd) {
+        out = sv0 ^ (sv << 1) ^ (sv << 2);
+    } else {
+        uint64_t a3 = sv & 0x3fffffffffffffffull;
+        out = sv0 ^ ((a3 << 1) | (sv_part >> 63)) ^
+              ((a3 << 2) | (sv_part >> 62));
+    }
+    *(uint64_t *)(a.sums + chain * 32 + 8 * li) = out;
 }
 
 /* ---- SHA-256 (one chain per lane), FIPS 180-4 -------------------------- */
@@ -480,7 +549,8 @@ hipError_t mec_launch_hash(int algo, const HashArgs *args,
         hipLaunchKernelGGL(sha256_batch_kernel, grid, blk, 0, stream, *args);
         break;
     case 2: /* HighwayHash256 */
-    case 3: /* HighwayHash256S */
+    case 3: /* HighwayHash256S: 4 lanes per chain */
+        grid.x = (uint32_t)((args->n_chains * 4 + 255) / 256);
         hipLaunchKernelGGL(hh256_batch_kernel, grid, blk, 0, stream, *args);
         break;
     case 4: /* BLAKE2b512 */

@@ -143,9 +143,12 @@ def test_stream_roundtrip_vs_oracle():
         assert streams == ostreams
         # full read
         assert e.decode_stream(streams, total, 0, total) == data
-        # ranged reads incl. cross-block
+        # ranged reads incl. cross-block and block-boundary-exact ends
+        # (the reference iterates into an empty end block and breaks,
+        #  cmd/erasure-decode.go:271-280)
         for off, ln in [(0, 1), (bs - 1, 2), (bs, bs), (100, 3 * bs),
-                        (total - 1, 1), (bs + 7, 2 * bs + 100)]:
+                        (total - 1, 1), (bs + 7, 2 * bs + 100),
+                        (0, bs), (0, 2 * bs), (bs, 2 * bs), (7, bs - 7)]:
             assert e.decode_stream(streams, total, off, ln) == data[off:off + ln]
         # with p drives missing
         dmg = list(streams)
