@@ -198,76 +198,96 @@ __device__ __forceinline__ uint64_t zip_odd(uint64_t v0, uint64_t v1) {
            ((v1 & 0xffull) << 48) | (v0 & 0xff00000000000000ull);
 }
 
-struct HH4 {
-    uint64_t v0, v1, mul0, mul1; /* this GPU lane's HighwayHash lane */
+struct HH2 {
+    /* this GPU lane's zipper PAIR of HighwayHash lanes: even lane owns HH
+     * lanes {0,1}, odd lane owns {2,3}.  The zipper merge operates on
+     * exactly these pairs, so the packet loop is fully lane-local — no
+     * cross-lane ops on the serial dependency chain. */
+    uint64_t v0[2], v1[2], mul0[2], mul1[2];
 };
 
 __device__ __forceinline__ uint64_t shfl_x(uint64_t v, int mask) {
     return __shfl_xor((unsigned long long)v, mask, 64);
 }
 
-__device__ __forceinline__ void hh4_update(HH4 &s, uint64_t lane_word,
-                                           bool odd) {
-    s.v1 += s.mul0 + lane_word;
-    s.mul0 ^= (s.v1 & 0xffffffffull) * (s.v0 >> 32);
-    s.v0 += s.mul1;
-    s.mul1 ^= (s.v0 & 0xffffffffull) * (s.v1 >> 32);
-    uint64_t pv1 = shfl_x(s.v1, 1);
-    s.v0 += odd ? zip_odd(pv1, s.v1) : zip_even(s.v1, pv1);
-    uint64_t pv0 = shfl_x(s.v0, 1);
-    s.v1 += odd ? zip_odd(pv0, s.v0) : zip_even(s.v0, pv0);
+__device__ __forceinline__ void hh2_update(HH2 &s, uint64_t w0, uint64_t w1) {
+    uint64_t w[2] = {w0, w1};
+#pragma unroll
+    for (int j = 0; j < 2; j++) {
+        s.v1[j] += s.mul0[j] + w[j];
+        s.mul0[j] ^= (s.v1[j] & 0xffffffffull) * (s.v0[j] >> 32);
+        s.v0[j] += s.mul1[j];
+        s.mul1[j] ^= (s.v0[j] & 0xffffffffull) * (s.v1[j] >> 32);
+    }
+    /* ZipperMergeAndAdd(v1[hi], v1[lo], &v0[hi], &v0[lo]) — pair-local */
+    uint64_t t0 = zip_even(s.v1[0], s.v1[1]);
+    uint64_t t1 = zip_odd(s.v1[0], s.v1[1]);
+    s.v0[0] += t0;
+    s.v0[1] += t1;
+    uint64_t u0 = zip_even(s.v0[0], s.v0[1]);
+    uint64_t u1 = zip_odd(s.v0[0], s.v0[1]);
+    s.v1[0] += u0;
+    s.v1[1] += u1;
 }
 
 __global__ void __launch_bounds__(256) hh256_batch_kernel(HashArgs a) {
     const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    const int64_t chain = tid >> 2;
-    const int li = (int)(tid & 3); /* HighwayHash lane index */
+    const int64_t chain = tid >> 1;
+    const int h = (int)(tid & 1); /* 0: HH lanes {0,1}; 1: HH lanes {2,3} */
     if (chain >= a.n_chains) return;
     const uint8_t *msg = chain_ptr(a, chain);
-    const bool odd = li & 1;
 
     const uint64_t init0[4] = {0xdbe6d5d5fe4cce2full, 0xa4093822299f31d0ull,
                                0x13198a2e03707344ull, 0x243f6a8885a308d3ull};
     const uint64_t init1[4] = {0x3bd39e10cb0ef593ull, 0xc0acf169b5f18a8cull,
                                0xbe5466cf34e90c6cull, 0x452821e638d01377ull};
-    HH4 s;
-    s.mul0 = init0[li];
-    s.mul1 = init1[li];
-    s.v0 = init0[li] ^ a.key[li];
-    s.v1 = init1[li] ^ ((a.key[li] >> 32) | (a.key[li] << 32));
+    HH2 s;
+#pragma unroll
+    for (int j = 0; j < 2; j++) {
+        int li = 2 * h + j;
+        s.mul0[j] = init0[li];
+        s.mul1[j] = init1[li];
+        s.v0[j] = init0[li] ^ a.key[li];
+        s.v1[j] = init1[li] ^ ((a.key[li] >> 32) | (a.key[li] << 32));
+    }
 
     int64_t len = a.msg_len;
-    const uint8_t *mp = msg + 8 * li;
+    const uint8_t *mp = msg + 16 * h; /* this lane's 16 B of each packet */
+#define HH2_LOAD_UPDATE(off)                                                 \
+    {                                                                        \
+        uint4 q = *(const uint4 *)(mp + (off));                              \
+        hh2_update(s, (uint64_t)q.x | ((uint64_t)q.y << 32),                 \
+                   (uint64_t)q.z | ((uint64_t)q.w << 32));                   \
+    }
     /* unrolled by 4 packets for memory-level parallelism */
     while (len >= 128) {
-        uint64_t w0 = *(const uint64_t *)(mp + 0);
-        uint64_t w1 = *(const uint64_t *)(mp + 32);
-        uint64_t w2 = *(const uint64_t *)(mp + 64);
-        uint64_t w3 = *(const uint64_t *)(mp + 96);
-        hh4_update(s, w0, odd);
-        hh4_update(s, w1, odd);
-        hh4_update(s, w2, odd);
-        hh4_update(s, w3, odd);
+        HH2_LOAD_UPDATE(0)
+        HH2_LOAD_UPDATE(32)
+        HH2_LOAD_UPDATE(64)
+        HH2_LOAD_UPDATE(96)
         mp += 128;
         len -= 128;
     }
     while (len >= 32) {
-        hh4_update(s, *(const uint64_t *)mp, odd);
+        HH2_LOAD_UPDATE(0)
         mp += 32;
         len -= 32;
     }
+#undef HH2_LOAD_UPDATE
     if (len > 0) {
         /* UpdateRemainder (published portable semantics); each lane builds
          * the full 32-byte packet locally — tail-only cost */
         const uint8_t *tail_msg = msg + (a.msg_len - len);
         const int mod32 = (int)len;
         const int mod4 = mod32 & 3;
-        s.v0 += ((uint64_t)mod32 << 32) + (uint64_t)mod32;
-        {
-            uint32_t h0 = (uint32_t)s.v1;
-            uint32_t h1 = (uint32_t)(s.v1 >> 32);
-            s.v1 = (uint32_t)((h0 << mod32) | (h0 >> (32 - mod32)));
-            s.v1 |= (uint64_t)((h1 << mod32) | (h1 >> (32 - mod32))) << 32;
+#pragma unroll
+        for (int j = 0; j < 2; j++) {
+            s.v0[j] += ((uint64_t)mod32 << 32) + (uint64_t)mod32;
+            uint32_t h0 = (uint32_t)s.v1[j];
+            uint32_t h1 = (uint32_t)(s.v1[j] >> 32);
+            s.v1[j] = (uint32_t)((h0 << mod32) | (h0 >> (32 - mod32)));
+            s.v1[j] |=
+                (uint64_t)((h1 << mod32) | (h1 >> (32 - mod32))) << 32;
         }
         uint8_t packet[32];
 #pragma unroll
@@ -281,31 +301,37 @@ __global__ void __launch_bounds__(256) hh256_batch_kernel(HashArgs a) {
             packet[17] = rem[mod4 >> 1];
             packet[18] = rem[mod4 - 1];
         }
-        uint64_t w = 0;
-        for (int bt = 7; bt >= 0; bt--) w = (w << 8) | packet[8 * li + bt];
-        hh4_update(s, w, odd);
+        uint64_t w[2];
+#pragma unroll
+        for (int j = 0; j < 2; j++) {
+            uint64_t v = 0;
+            for (int bt = 7; bt >= 0; bt--)
+                v = (v << 8) | packet[16 * h + 8 * j + bt];
+            w[j] = v;
+        }
+        hh2_update(s, w[0], w[1]);
     }
-    /* finalization: 10 permute-update rounds; permuted lane li reads
-     * rot32(v0[li ^ 2]) */
+    /* finalization: 10 permute-update rounds.  Permuted word for HH lane l
+     * is rot32(v0[l ^ 2]) — the partner lane's same-position word. */
 #pragma unroll 1
     for (int r = 0; r < 10; r++) {
-        uint64_t pv = shfl_x(s.v0, 2);
-        hh4_update(s, (pv >> 32) | (pv << 32), odd);
+        uint64_t p0 = shfl_x(s.v0[0], 1);
+        uint64_t p1 = shfl_x(s.v0[1], 1);
+        hh2_update(s, (p0 >> 32) | (p0 << 32), (p1 >> 32) | (p1 << 32));
     }
-    /* modular reduction: even lanes all-local; odd lanes need partner's
-     * (v1+mul1) */
-    uint64_t sv = s.v1 + s.mul1;
-    uint64_t sv0 = s.v0 + s.mul0;
-    uint64_t sv_part = shfl_x(sv, 1);
-    uint64_t out;
-    if (!odd) {
-        out = sv0 ^ (sv << 1) ^ (sv << 2);
-    } else {
-        uint64_t a3 = sv & 0x3fffffffffffffffull;
-        out = sv0 ^ ((a3 << 1) | (sv_part >> 63)) ^
-              ((a3 << 2) | (sv_part >> 62));
-    }
-    *(uint64_t *)(a.sums + chain * 32 + 8 * li) = out;
+    /* modular reduction — fully pair-local: even lane emits hash[0..1],
+     * odd lane hash[2..3] */
+    uint64_t a2 = s.v1[0] + s.mul1[0];
+    uint64_t a3 = (s.v1[1] + s.mul1[1]) & 0x3fffffffffffffffull;
+    uint64_t o0 = (s.v0[0] + s.mul0[0]) ^ (a2 << 1) ^ (a2 << 2);
+    uint64_t o1 = (s.v0[1] + s.mul0[1]) ^ ((a3 << 1) | (a2 >> 63)) ^
+                  ((a3 << 2) | (a2 >> 62));
+    uint4 out;
+    out.x = (uint32_t)o0;
+    out.y = (uint32_t)(o0 >> 32);
+    out.z = (uint32_t)o1;
+    out.w = (uint32_t)(o1 >> 32);
+    *(uint4 *)(a.sums + chain * 32 + 16 * h) = out;
 }
 
 /* ---- SHA-256 (one chain per lane), FIPS 180-4 -------------------------- */
@@ -549,8 +575,8 @@ hipError_t mec_launch_hash(int algo, const HashArgs *args,
         hipLaunchKernelGGL(sha256_batch_kernel, grid, blk, 0, stream, *args);
         break;
     case 2: /* HighwayHash256 */
-    case 3: /* HighwayHash256S: 4 lanes per chain */
-        grid.x = (uint32_t)((args->n_chains * 4 + 255) / 256);
+    case 3: /* HighwayHash256S: 2 lanes per chain (zipper pairs) */
+        grid.x = (uint32_t)((args->n_chains * 2 + 255) / 256);
         hipLaunchKernelGGL(hh256_batch_kernel, grid, blk, 0, stream, *args);
         break;
     case 4: /* BLAKE2b512 */
