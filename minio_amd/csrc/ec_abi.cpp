@@ -227,7 +227,22 @@ static mec_status encode_dev_locked(mec_ctx *ctx, int n, const void *data_dev,
     /* per-call shard size mirrors Split: ceil(block_len/d)
      * (cmd/erasure-coding.go:81 + :117); rows stay at ctx->stride */
     const int64_t S_call = ceil_frac(block_len, d);
-    /* GF parity rows, in groups of <= MEC_KMAX_E */
+    /* specialized straight-line kernel for common geometries */
+    {
+        GfEncArgs ea{};
+        ea.data = (const uint8_t *)data_dev;
+        ea.parity = (uint8_t *)parity_dev;
+        ea.row_stride = ctx->stride;
+        ea.shard_len = S_call;
+        hipError_t he =
+            mec_launch_gf_encode_spec(d, p, &ea, n, ctx->stream);
+        if (he == hipSuccess) goto gf_done;
+        if (he != hipErrorNotSupported) {
+            set_err("gf_encode_spec", he);
+            return MEC_ERR_HIP;
+        }
+    }
+    /* generic fallback: GF parity rows, in groups of <= MEC_KMAX_E */
     for (int i0 = 0; i0 < p; i0 += MEC_KMAX_E) {
         int e = p - i0 > MEC_KMAX_E ? MEC_KMAX_E : p - i0;
         GfMatmulArgs a{};
@@ -246,6 +261,7 @@ static mec_status encode_dev_locked(mec_ctx *ctx, int n, const void *data_dev,
                     ctx->enc_matrix[(size_t)(d + i0 + i) * d + k];
         HIP_TRY(mec_launch_gf_matmul(&a, e, n, ctx->stream));
     }
+gf_done:
     if (sums_dev != nullptr) {
         if (!hash_size(algo)) return MEC_ERR_INVALID_ARG;
         HashArgs h{};

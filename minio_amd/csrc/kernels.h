@@ -38,9 +38,19 @@ struct HashArgs {
     uint64_t key[4];       /* HighwayHash key (little-endian words) */
 };
 
+/* Specialized-encode args (constexpr-matrix kernels) */
+struct GfEncArgs {
+    const uint8_t *data;
+    uint8_t *parity;
+    int64_t row_stride;
+    int64_t shard_len;
+};
+
 extern "C" {
 hipError_t mec_launch_gf_matmul(const GfMatmulArgs *args, int n_dst, int n,
                                 hipStream_t stream);
+hipError_t mec_launch_gf_encode_spec(int d, int p, const GfEncArgs *args,
+                                     int n, hipStream_t stream);
 hipError_t mec_launch_hash(int algo, const HashArgs *args, hipStream_t stream);
 }
 

@@ -45,6 +45,50 @@ __device__ __forceinline__ void xor4(uint4 &a, const uint4 &b) {
     a.w ^= b.w;
 }
 
+/* ---- specialized encode kernels (constexpr matrix) ---------------------
+ *
+ * For the common geometries the encode matrix is baked in at compile time
+ * (ec_matrices_gen.h, generated from the product's own matrix builder), so
+ * the inner loop is a straight-line XOR schedule: no per-bit branches (the
+ * generic kernel's 60% front-end issue-stall), no coefficient loads, and
+ * doubling steps stop at each column's highest set bit. */
+#include "ec_matrices_gen.h"
+
+template <int D, int P, const uint8_t (&MAT)[P][D]>
+__global__ void __launch_bounds__(256) gf_encode_kernel(GfEncArgs a) {
+    const int b = blockIdx.y;
+    const int64_t cols = (a.shard_len + 15) >> 4;
+    const uint8_t *__restrict__ sbase = a.data + (int64_t)b * D * a.row_stride;
+    uint8_t *__restrict__ obase = a.parity + (int64_t)b * P * a.row_stride;
+
+    for (int64_t c = blockIdx.x * blockDim.x + threadIdx.x; c < cols;
+         c += (int64_t)gridDim.x * blockDim.x) {
+        const int64_t j = c << 4;
+        uint4 acc[P];
+#pragma unroll
+        for (int i = 0; i < P; i++) acc[i] = uint4{0, 0, 0, 0};
+#pragma unroll
+        for (int k = 0; k < D; k++) {
+            uint4 pw = *(const uint4 *)(sbase + (int64_t)k * a.row_stride + j);
+#pragma unroll
+            for (int bit = 0; bit < 8; bit++) {
+                uint32_t need = 0;
+#pragma unroll
+                for (int i = 0; i < P; i++)
+                    need |= (uint32_t)MAT[i][k] >> bit;
+                if (!need) break; /* compile-time folded */
+                if (bit) gf2x4(pw);
+#pragma unroll
+                for (int i = 0; i < P; i++)
+                    if ((MAT[i][k] >> bit) & 1) xor4(acc[i], pw);
+            }
+        }
+#pragma unroll
+        for (int i = 0; i < P; i++)
+            *(uint4 *)(obase + (int64_t)i * a.row_stride + j) = acc[i];
+    }
+}
+
 /* ---- generic GF matrix-multiply over shard rows ------------------------
  *
  * out[t][j] = sum_k mat[t][k] * src[k][j]  (GF(2^8)), per batch item.
@@ -259,14 +303,19 @@ __global__ void __launch_bounds__(256) hh256_batch_kernel(HashArgs a) {
         hh2_update(s, (uint64_t)q.x | ((uint64_t)q.y << 32),                 \
                    (uint64_t)q.z | ((uint64_t)q.w << 32));                   \
     }
-    /* unrolled by 4 packets for memory-level parallelism */
-    while (len >= 128) {
-        HH2_LOAD_UPDATE(0)
-        HH2_LOAD_UPDATE(32)
-        HH2_LOAD_UPDATE(64)
-        HH2_LOAD_UPDATE(96)
-        mp += 128;
-        len -= 128;
+    /* 16-packet prefetch: the hash chain is serial, so the only latency
+     * cover is load-ahead depth (profile r01: 43% of wave-cycles were
+     * memory waits at depth 4) */
+    while (len >= 512) {
+        uint4 q[16];
+#pragma unroll
+        for (int t = 0; t < 16; t++) q[t] = *(const uint4 *)(mp + 32 * t);
+#pragma unroll
+        for (int t = 0; t < 16; t++)
+            hh2_update(s, (uint64_t)q[t].x | ((uint64_t)q[t].y << 32),
+                       (uint64_t)q[t].z | ((uint64_t)q[t].w << 32));
+        mp += 512;
+        len -= 512;
     }
     while (len >= 32) {
         HH2_LOAD_UPDATE(0)
@@ -540,6 +589,28 @@ __global__ void __launch_bounds__(256) blake2b512_batch_kernel(HashArgs a) {
 /* ---- launch wrappers (called from ec_abi.cpp) -------------------------- */
 
 extern "C" {
+
+/* Specialized encode launch; returns hipErrorNotSupported when (d,p) has no
+ * compiled specialization (caller falls back to the generic kernel). */
+hipError_t mec_launch_gf_encode_spec(int d, int p, const GfEncArgs *args,
+                                     int n, hipStream_t stream) {
+    const int64_t cols = (args->shard_len + 15) >> 4;
+    int64_t max_x = (cols + 255) / 256;
+    int64_t want_x = (2048 + n - 1) / n;
+    int64_t blocks_x = want_x < max_x ? want_x : max_x;
+    if (blocks_x < 1) blocks_x = 1;
+    dim3 grid((uint32_t)blocks_x, n);
+    dim3 blk(256);
+#define X(D, P)                                                              \
+    if (d == D && p == P) {                                                  \
+        hipLaunchKernelGGL((gf_encode_kernel<D, P, MAT_##D##_##P>), grid,    \
+                           blk, 0, stream, *args);                           \
+        return hipGetLastError();                                            \
+    }
+    MEC_SPECIALIZED_GEOS(X)
+#undef X
+    return hipErrorNotSupported;
+}
 
 hipError_t mec_launch_gf_matmul(const GfMatmulArgs *args, int n_dst, int n,
                                 hipStream_t stream) {
