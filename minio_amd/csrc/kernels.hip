@@ -226,20 +226,32 @@ __device__ __forceinline__ const uint8_t *chain_ptr(const HashArgs &a,
  * hashing is sequential per shard, so chains x lanes is the only
  * parallelism available (SURVEY.md §7 hard part (b)). */
 
-__device__ __forceinline__ uint64_t zip_even(uint64_t v0, uint64_t v1) {
-    /* add0 expression of ZipperMergeAndAdd(v1=v1[odd], v0=v1[even], ...) */
-    return (((v0 & 0xff000000ull) | (v1 & 0xff00000000ull)) >> 24) |
-           (((v0 & 0xff0000000000ull) | (v1 & 0xff000000000000ull)) >> 16) |
-           (v0 & 0xff0000ull) | ((v0 & 0xff00ull) << 32) |
-           ((v1 & 0xff00000000000000ull) >> 8) | (v0 << 56);
+/* The zipper merge is a pure byte permutation of the pair (the reference
+ * implements it with pshufb); on CDNA4 that is v_perm_b32: 3 perms + 1 or
+ * per output word instead of a ~17-op shift/mask tree.
+ *   zip_even out bytes (LSB..MSB): [A3 B4 A2 A5 | B6 A1 B7 A0]
+ *   zip_odd  out bytes:            [B3 A4 B2 B5 | B1 A6 B0 A7]
+ * where A = v1[even], B = v1[odd] of the pair.  v_perm pool = {hi:lo},
+ * selector byte 12 yields 0x00. */
+__device__ __forceinline__ uint32_t permb(uint32_t hi, uint32_t lo,
+                                          uint32_t sel) {
+    return __builtin_amdgcn_perm(hi, lo, sel);
 }
 
-__device__ __forceinline__ uint64_t zip_odd(uint64_t v0, uint64_t v1) {
-    /* add1 expression; v0 = v1[even] (partner), v1 = v1[odd] (own) */
-    return (((v1 & 0xff000000ull) | (v0 & 0xff00000000ull)) >> 24) |
-           (v1 & 0xff0000ull) | ((v1 & 0xff0000000000ull) >> 16) |
-           ((v1 & 0xff00ull) << 24) | ((v0 & 0xff000000000000ull) >> 8) |
-           ((v1 & 0xffull) << 48) | (v0 & 0xff00000000000000ull);
+__device__ __forceinline__ uint64_t zip_even(uint64_t A, uint64_t B) {
+    uint32_t a_lo = (uint32_t)A, a_hi = (uint32_t)(A >> 32);
+    uint32_t b_hi = (uint32_t)(B >> 32);
+    uint32_t lo = permb(a_hi, a_lo, 0x05020C03u) | permb(0u, b_hi, 0x0C0C000Cu);
+    uint32_t hi = permb(b_hi, a_lo, 0x00070106u);
+    return ((uint64_t)hi << 32) | lo;
+}
+
+__device__ __forceinline__ uint64_t zip_odd(uint64_t A, uint64_t B) {
+    uint32_t a_hi = (uint32_t)(A >> 32);
+    uint32_t b_lo = (uint32_t)B, b_hi = (uint32_t)(B >> 32);
+    uint32_t lo = permb(b_hi, b_lo, 0x05020C03u) | permb(0u, a_hi, 0x0C0C000Cu);
+    uint32_t hi = permb(a_hi, b_lo, 0x07000601u);
+    return ((uint64_t)hi << 32) | lo;
 }
 
 struct HH2 {
