@@ -75,7 +75,9 @@ struct mec_ctx {
     int64_t stride = 0; /* device row stride = align64(S) */
     std::vector<uint8_t> enc_matrix; /* (d+p) x d */
     hipStream_t stream = nullptr;
+    hipStream_t stream2 = nullptr; /* hash(data) overlap lane */
     hipEvent_t ev_start = nullptr, ev_stop = nullptr;
+    hipEvent_t ev_gf = nullptr, ev_h = nullptr, ev_fork = nullptr;
     std::mutex mu;
 
     /* grow-only scratch (device + pinned host) for host-pointer calls */
@@ -186,8 +188,12 @@ mec_status mec_ctx_create(int d, int p, int64_t block_size, int device,
     }
     hipError_t e = hipSetDevice(device);
     if (e == hipSuccess) e = hipStreamCreate(&ctx->stream);
+    if (e == hipSuccess) e = hipStreamCreate(&ctx->stream2);
     if (e == hipSuccess) e = hipEventCreate(&ctx->ev_start);
     if (e == hipSuccess) e = hipEventCreate(&ctx->ev_stop);
+    if (e == hipSuccess) e = hipEventCreateWithFlags(&ctx->ev_gf, hipEventDisableTiming);
+    if (e == hipSuccess) e = hipEventCreateWithFlags(&ctx->ev_h, hipEventDisableTiming);
+    if (e == hipSuccess) e = hipEventCreateWithFlags(&ctx->ev_fork, hipEventDisableTiming);
     if (e != hipSuccess) {
         set_err("ctx_create", e);
         delete ctx;
@@ -210,6 +216,10 @@ void mec_ctx_destroy(mec_ctx *ctx) {
     if (ctx->pin) (void)hipHostFree(ctx->pin);
     if (ctx->ev_start) (void)hipEventDestroy(ctx->ev_start);
     if (ctx->ev_stop) (void)hipEventDestroy(ctx->ev_stop);
+    if (ctx->ev_gf) (void)hipEventDestroy(ctx->ev_gf);
+    if (ctx->ev_h) (void)hipEventDestroy(ctx->ev_h);
+    if (ctx->ev_fork) (void)hipEventDestroy(ctx->ev_fork);
+    if (ctx->stream2) (void)hipStreamDestroy(ctx->stream2);
     if (ctx->stream) (void)hipStreamDestroy(ctx->stream);
     delete ctx;
 }
@@ -227,6 +237,10 @@ static mec_status encode_dev_locked(mec_ctx *ctx, int n, const void *data_dev,
     /* per-call shard size mirrors Split: ceil(block_len/d)
      * (cmd/erasure-coding.go:81 + :117); rows stay at ctx->stride */
     const int64_t S_call = ceil_frac(block_len, d);
+    /* fork point: stream2 work (concurrent data-hash) must order after
+     * everything already enqueued on the main stream (e.g. H2D staging) */
+    if (sums_dev != nullptr)
+        HIP_TRY(hipEventRecord(ctx->ev_fork, ctx->stream));
     /* specialized straight-line kernel for common geometries */
     {
         GfEncArgs ea{};
@@ -264,17 +278,32 @@ static mec_status encode_dev_locked(mec_ctx *ctx, int n, const void *data_dev,
 gf_done:
     if (sums_dev != nullptr) {
         if (!hash_size(algo)) return MEC_ERR_INVALID_ARG;
+        /* overlap: hash(data) on stream2 runs concurrently with the GF
+         * kernel (it only reads the input); hash(parity) follows the GF
+         * completion event; the main stream then joins, so callers see one
+         * fused synchronous operation. */
         HashArgs h{};
         h.data = (const uint8_t *)data_dev;
         h.parity = (const uint8_t *)parity_dev;
         h.sums = (uint8_t *)sums_dev;
         h.row_stride = ctx->stride;
         h.msg_len = S_call;
-        h.n_chains = (int64_t)n * (d + p);
         h.d = d;
         h.p = p;
         memcpy(h.key, kMagicHHKey, 32);
-        HIP_TRY(mec_launch_hash(algo, &h, ctx->stream));
+        /* stream2 must not start before prior work on the main stream
+         * (e.g. the H2D staging copy) has completed */
+        HIP_TRY(hipEventRecord(ctx->ev_gf, ctx->stream)); /* after GF */
+        HIP_TRY(hipStreamWaitEvent(ctx->stream2, ctx->ev_fork, 0));
+        h.mode = MEC_HASH_DATA;
+        h.n_chains = (int64_t)n * d;
+        HIP_TRY(mec_launch_hash(algo, &h, ctx->stream2));
+        HIP_TRY(hipStreamWaitEvent(ctx->stream2, ctx->ev_gf, 0));
+        h.mode = MEC_HASH_PARITY;
+        h.n_chains = (int64_t)n * p;
+        HIP_TRY(mec_launch_hash(algo, &h, ctx->stream2));
+        HIP_TRY(hipEventRecord(ctx->ev_h, ctx->stream2));
+        HIP_TRY(hipStreamWaitEvent(ctx->stream, ctx->ev_h, 0));
     }
     return MEC_OK;
 }

@@ -26,15 +26,21 @@ struct GfMatmulArgs {
     uint8_t mat[MEC_KMAX_E * MEC_KMAX_D]; /* row t: coefficients over src */
 };
 
+/* mode: which shard rows this launch hashes.  Sums always land in the
+ * fused n x (d+p) x digest layout so a data-only and a parity-only launch
+ * together produce exactly the single-launch result. */
+enum MecHashMode { MEC_HASH_ALL = 0, MEC_HASH_DATA = 1, MEC_HASH_PARITY = 2 };
+
 struct HashArgs {
     const uint8_t *data;   /* data-shard rows, n*d*row_stride */
     const uint8_t *parity; /* parity rows, n*p*row_stride; NULL => simple
                               strided layout: chain i at data+i*row_stride */
-    uint8_t *sums;         /* n_chains * digest_size */
+    uint8_t *sums;         /* n * (d+p) * digest_size (fused layout) */
     int64_t row_stride;
     int64_t msg_len;       /* bytes hashed per chain (the padded shard) */
-    int64_t n_chains;      /* n * (d+p), or n for simple layout */
+    int64_t n_chains;      /* chains in THIS launch: n*(d+p), n*d or n*p */
     int d, p;
+    int mode;              /* MecHashMode */
     uint64_t key[4];       /* HighwayHash key (little-endian words) */
 };
 

@@ -206,14 +206,32 @@ __device__ __forceinline__ void hh_finalize256(HHState &s, uint64_t out[4]) {
     out[2] = (s.v0[2] + s.mul0[2]) ^ (a2 << 1) ^ (a2 << 2);
 }
 
-/* Map chain index -> shard pointer for the fused encode layout:
- * chains are (block b, shard s); s < d lives in `data`, else in `parity`. */
+/* Map chain index -> (shard pointer, sum slot) for the fused encode
+ * layout.  Sums always use the n x (d+p) layout regardless of mode, so
+ * data-only + parity-only launches compose to the single-launch result. */
 __device__ __forceinline__ const uint8_t *chain_ptr(const HashArgs &a,
-                                                    int64_t chain) {
-    if (a.parity == nullptr) return a.data + chain * a.row_stride;
+                                                    int64_t chain,
+                                                    int64_t &sum_idx) {
+    if (a.parity == nullptr) {
+        sum_idx = chain;
+        return a.data + chain * a.row_stride;
+    }
     const int total = a.d + a.p;
+    if (a.mode == MEC_HASH_DATA) {
+        const int64_t b = chain / a.d;
+        const int s = (int)(chain % a.d);
+        sum_idx = b * total + s;
+        return a.data + (b * a.d + s) * a.row_stride;
+    }
+    if (a.mode == MEC_HASH_PARITY) {
+        const int64_t b = chain / a.p;
+        const int s = (int)(chain % a.p);
+        sum_idx = b * total + a.d + s;
+        return a.parity + (b * a.p + s) * a.row_stride;
+    }
     const int64_t b = chain / total;
     const int s = (int)(chain % total);
+    sum_idx = chain;
     if (s < a.d)
         return a.data + (b * a.d + s) * a.row_stride;
     return a.parity + (b * a.p + (s - a.d)) * a.row_stride;
@@ -291,7 +309,8 @@ __global__ void __launch_bounds__(256) hh256_batch_kernel(HashArgs a) {
     const int64_t chain = tid >> 1;
     const int h = (int)(tid & 1); /* 0: HH lanes {0,1}; 1: HH lanes {2,3} */
     if (chain >= a.n_chains) return;
-    const uint8_t *msg = chain_ptr(a, chain);
+    int64_t sum_idx;
+    const uint8_t *msg = chain_ptr(a, chain, sum_idx);
 
     const uint64_t init0[4] = {0xdbe6d5d5fe4cce2full, 0xa4093822299f31d0ull,
                                0x13198a2e03707344ull, 0x243f6a8885a308d3ull};
@@ -392,7 +411,7 @@ __global__ void __launch_bounds__(256) hh256_batch_kernel(HashArgs a) {
     out.y = (uint32_t)(o0 >> 32);
     out.z = (uint32_t)o1;
     out.w = (uint32_t)(o1 >> 32);
-    *(uint4 *)(a.sums + chain * 32 + 16 * h) = out;
+    *(uint4 *)(a.sums + sum_idx * 32 + 16 * h) = out;
 }
 
 /* ---- SHA-256 (one chain per lane), FIPS 180-4 -------------------------- */
@@ -452,7 +471,8 @@ __device__ void sha256_block(uint32_t h[8], const uint32_t w_in[16]) {
 __global__ void __launch_bounds__(256) sha256_batch_kernel(HashArgs a) {
     const int64_t chain = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (chain >= a.n_chains) return;
-    const uint8_t *msg = chain_ptr(a, chain);
+    int64_t sum_idx;
+    const uint8_t *msg = chain_ptr(a, chain, sum_idx);
 
     uint32_t h[8] = {0x6a09e667, 0xbb67ae85, 0x3c6ef372, 0xa54ff53a,
                      0x510e527f, 0x9b05688c, 0x1f83d9ab, 0x5be0cd19};
@@ -492,7 +512,7 @@ __global__ void __launch_bounds__(256) sha256_batch_kernel(HashArgs a) {
             sha256_block(h, w);
         }
     }
-    uint8_t *out = a.sums + chain * 32;
+    uint8_t *out = a.sums + sum_idx * 32;
     for (int i = 0; i < 8; i++)
         *(uint32_t *)(out + 4 * i) = bswap32(h[i]);
 }
@@ -562,7 +582,8 @@ __device__ void b2b_compress(uint64_t h[8], const uint64_t m[16], uint64_t t,
 __global__ void __launch_bounds__(256) blake2b512_batch_kernel(HashArgs a) {
     const int64_t chain = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (chain >= a.n_chains) return;
-    const uint8_t *msg = chain_ptr(a, chain);
+    int64_t sum_idx;
+    const uint8_t *msg = chain_ptr(a, chain, sum_idx);
 
     uint64_t h[8];
 #pragma unroll
@@ -594,7 +615,7 @@ __global__ void __launch_bounds__(256) blake2b512_batch_kernel(HashArgs a) {
         }
         b2b_compress(h, m, (uint64_t)a.msg_len, true);
     }
-    uint8_t *out = a.sums + chain * 64;
+    uint8_t *out = a.sums + sum_idx * 64;
     for (int i = 0; i < 8; i++) *(uint64_t *)(out + 8 * i) = h[i];
 }
 
