@@ -304,114 +304,143 @@ __device__ __forceinline__ void hh2_update(HH2 &s, uint64_t w0, uint64_t w1) {
     s.v1[1] += u1;
 }
 
+/* NC = independent chains per lane (ILP): the hash chain is serial, so a
+ * single chain per lane leaves every dependent-op latency exposed (measured
+ * ~340 cyc/packet vs ~104 issue-bound).  Interleaving NC independent
+ * chains' packet updates in one lane fills those stalls. */
+template <int NC>
 __global__ void __launch_bounds__(256) hh256_batch_kernel(HashArgs a) {
     const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    const int64_t chain = tid >> 1;
+    const int64_t slot = tid >> 1;
     const int h = (int)(tid & 1); /* 0: HH lanes {0,1}; 1: HH lanes {2,3} */
-    if (chain >= a.n_chains) return;
-    int64_t sum_idx;
-    const uint8_t *msg = chain_ptr(a, chain, sum_idx);
+    const int64_t c0 = slot * NC;
+    if (c0 >= a.n_chains) return;
+
+    bool act[NC];
+    const uint8_t *mp[NC];
+    int64_t sum_idx[NC];
+#pragma unroll
+    for (int u = 0; u < NC; u++) {
+        act[u] = c0 + u < a.n_chains;
+        mp[u] = chain_ptr(a, act[u] ? c0 + u : c0, sum_idx[u]) + 16 * h;
+    }
 
     const uint64_t init0[4] = {0xdbe6d5d5fe4cce2full, 0xa4093822299f31d0ull,
                                0x13198a2e03707344ull, 0x243f6a8885a308d3ull};
     const uint64_t init1[4] = {0x3bd39e10cb0ef593ull, 0xc0acf169b5f18a8cull,
                                0xbe5466cf34e90c6cull, 0x452821e638d01377ull};
-    HH2 s;
+    HH2 s[NC];
 #pragma unroll
-    for (int j = 0; j < 2; j++) {
-        int li = 2 * h + j;
-        s.mul0[j] = init0[li];
-        s.mul1[j] = init1[li];
-        s.v0[j] = init0[li] ^ a.key[li];
-        s.v1[j] = init1[li] ^ ((a.key[li] >> 32) | (a.key[li] << 32));
-    }
+    for (int u = 0; u < NC; u++)
+#pragma unroll
+        for (int j = 0; j < 2; j++) {
+            int li = 2 * h + j;
+            s[u].mul0[j] = init0[li];
+            s[u].mul1[j] = init1[li];
+            s[u].v0[j] = init0[li] ^ a.key[li];
+            s[u].v1[j] = init1[li] ^ ((a.key[li] >> 32) | (a.key[li] << 32));
+        }
 
     int64_t len = a.msg_len;
-    const uint8_t *mp = msg + 16 * h; /* this lane's 16 B of each packet */
-#define HH2_LOAD_UPDATE(off)                                                 \
-    {                                                                        \
-        uint4 q = *(const uint4 *)(mp + (off));                              \
-        hh2_update(s, (uint64_t)q.x | ((uint64_t)q.y << 32),                 \
-                   (uint64_t)q.z | ((uint64_t)q.w << 32));                   \
-    }
-    /* 16-packet prefetch: the hash chain is serial, so the only latency
-     * cover is load-ahead depth (profile r01: 43% of wave-cycles were
-     * memory waits at depth 4) */
-    while (len >= 512) {
-        uint4 q[16];
+    /* 8-packet prefetch per chain, NC chains interleaved */
+    while (len >= 256) {
+        uint4 q[NC][8];
 #pragma unroll
-        for (int t = 0; t < 16; t++) q[t] = *(const uint4 *)(mp + 32 * t);
+        for (int t = 0; t < 8; t++)
 #pragma unroll
-        for (int t = 0; t < 16; t++)
-            hh2_update(s, (uint64_t)q[t].x | ((uint64_t)q[t].y << 32),
-                       (uint64_t)q[t].z | ((uint64_t)q[t].w << 32));
-        mp += 512;
-        len -= 512;
+            for (int u = 0; u < NC; u++)
+                q[u][t] = *(const uint4 *)(mp[u] + 32 * t);
+#pragma unroll
+        for (int t = 0; t < 8; t++)
+#pragma unroll
+            for (int u = 0; u < NC; u++)
+                hh2_update(s[u],
+                           (uint64_t)q[u][t].x | ((uint64_t)q[u][t].y << 32),
+                           (uint64_t)q[u][t].z | ((uint64_t)q[u][t].w << 32));
+#pragma unroll
+        for (int u = 0; u < NC; u++) mp[u] += 256;
+        len -= 256;
     }
     while (len >= 32) {
-        HH2_LOAD_UPDATE(0)
-        mp += 32;
+#pragma unroll
+        for (int u = 0; u < NC; u++) {
+            uint4 q = *(const uint4 *)mp[u];
+            hh2_update(s[u], (uint64_t)q.x | ((uint64_t)q.y << 32),
+                       (uint64_t)q.z | ((uint64_t)q.w << 32));
+            mp[u] += 32;
+        }
         len -= 32;
     }
-#undef HH2_LOAD_UPDATE
     if (len > 0) {
-        /* UpdateRemainder (published portable semantics); each lane builds
-         * the full 32-byte packet locally — tail-only cost */
-        const uint8_t *tail_msg = msg + (a.msg_len - len);
+        /* UpdateRemainder (published portable semantics); tail-only cost */
         const int mod32 = (int)len;
         const int mod4 = mod32 & 3;
 #pragma unroll
-        for (int j = 0; j < 2; j++) {
-            s.v0[j] += ((uint64_t)mod32 << 32) + (uint64_t)mod32;
-            uint32_t h0 = (uint32_t)s.v1[j];
-            uint32_t h1 = (uint32_t)(s.v1[j] >> 32);
-            s.v1[j] = (uint32_t)((h0 << mod32) | (h0 >> (32 - mod32)));
-            s.v1[j] |=
-                (uint64_t)((h1 << mod32) | (h1 >> (32 - mod32))) << 32;
-        }
-        uint8_t packet[32];
+        for (int u = 0; u < NC; u++) {
+            const uint8_t *tail_msg = mp[u] - 16 * h; /* row + body offset */
 #pragma unroll
-        for (int i = 0; i < 32; i++) packet[i] = 0;
-        for (int i = 0; i < (mod32 & ~3); i++) packet[i] = tail_msg[i];
-        const uint8_t *rem = tail_msg + (mod32 & ~3);
-        if (mod32 & 16) {
-            for (int i = 0; i < 4; i++) packet[28 + i] = rem[i + mod4 - 4];
-        } else if (mod4) {
-            packet[16] = rem[0];
-            packet[17] = rem[mod4 >> 1];
-            packet[18] = rem[mod4 - 1];
-        }
-        uint64_t w[2];
+            for (int j = 0; j < 2; j++) {
+                s[u].v0[j] += ((uint64_t)mod32 << 32) + (uint64_t)mod32;
+                uint32_t h0 = (uint32_t)s[u].v1[j];
+                uint32_t h1 = (uint32_t)(s[u].v1[j] >> 32);
+                s[u].v1[j] =
+                    (uint32_t)((h0 << mod32) | (h0 >> (32 - mod32)));
+                s[u].v1[j] |=
+                    (uint64_t)((h1 << mod32) | (h1 >> (32 - mod32))) << 32;
+            }
+            uint8_t packet[32];
 #pragma unroll
-        for (int j = 0; j < 2; j++) {
-            uint64_t v = 0;
-            for (int bt = 7; bt >= 0; bt--)
-                v = (v << 8) | packet[16 * h + 8 * j + bt];
-            w[j] = v;
+            for (int i = 0; i < 32; i++) packet[i] = 0;
+            for (int i = 0; i < (mod32 & ~3); i++) packet[i] = tail_msg[i];
+            const uint8_t *rem = tail_msg + (mod32 & ~3);
+            if (mod32 & 16) {
+                for (int i = 0; i < 4; i++)
+                    packet[28 + i] = rem[i + mod4 - 4];
+            } else if (mod4) {
+                packet[16] = rem[0];
+                packet[17] = rem[mod4 >> 1];
+                packet[18] = rem[mod4 - 1];
+            }
+            uint64_t w[2];
+#pragma unroll
+            for (int j = 0; j < 2; j++) {
+                uint64_t v = 0;
+                for (int bt = 7; bt >= 0; bt--)
+                    v = (v << 8) | packet[16 * h + 8 * j + bt];
+                w[j] = v;
+            }
+            hh2_update(s[u], w[0], w[1]);
         }
-        hh2_update(s, w[0], w[1]);
     }
-    /* finalization: 10 permute-update rounds.  Permuted word for HH lane l
+    /* finalization: 10 permute-update rounds; permuted word for HH lane l
      * is rot32(v0[l ^ 2]) — the partner lane's same-position word. */
 #pragma unroll 1
     for (int r = 0; r < 10; r++) {
-        uint64_t p0 = shfl_x(s.v0[0], 1);
-        uint64_t p1 = shfl_x(s.v0[1], 1);
-        hh2_update(s, (p0 >> 32) | (p0 << 32), (p1 >> 32) | (p1 << 32));
+#pragma unroll
+        for (int u = 0; u < NC; u++) {
+            uint64_t p0 = shfl_x(s[u].v0[0], 1);
+            uint64_t p1 = shfl_x(s[u].v0[1], 1);
+            hh2_update(s[u], (p0 >> 32) | (p0 << 32),
+                       (p1 >> 32) | (p1 << 32));
+        }
     }
-    /* modular reduction — fully pair-local: even lane emits hash[0..1],
-     * odd lane hash[2..3] */
-    uint64_t a2 = s.v1[0] + s.mul1[0];
-    uint64_t a3 = (s.v1[1] + s.mul1[1]) & 0x3fffffffffffffffull;
-    uint64_t o0 = (s.v0[0] + s.mul0[0]) ^ (a2 << 1) ^ (a2 << 2);
-    uint64_t o1 = (s.v0[1] + s.mul0[1]) ^ ((a3 << 1) | (a2 >> 63)) ^
-                  ((a3 << 2) | (a2 >> 62));
-    uint4 out;
-    out.x = (uint32_t)o0;
-    out.y = (uint32_t)(o0 >> 32);
-    out.z = (uint32_t)o1;
-    out.w = (uint32_t)(o1 >> 32);
-    *(uint4 *)(a.sums + sum_idx * 32 + 16 * h) = out;
+    /* modular reduction — pair-local: even lane emits hash[0..1], odd lane
+     * hash[2..3] */
+#pragma unroll
+    for (int u = 0; u < NC; u++) {
+        if (!act[u]) continue;
+        uint64_t a2 = s[u].v1[0] + s[u].mul1[0];
+        uint64_t a3 = (s[u].v1[1] + s[u].mul1[1]) & 0x3fffffffffffffffull;
+        uint64_t o0 = (s[u].v0[0] + s[u].mul0[0]) ^ (a2 << 1) ^ (a2 << 2);
+        uint64_t o1 = (s[u].v0[1] + s[u].mul0[1]) ^
+                      ((a3 << 1) | (a2 >> 63)) ^ ((a3 << 2) | (a2 >> 62));
+        uint4 out;
+        out.x = (uint32_t)o0;
+        out.y = (uint32_t)(o0 >> 32);
+        out.z = (uint32_t)o1;
+        out.w = (uint32_t)(o1 >> 32);
+        *(uint4 *)(a.sums + sum_idx[u] * 32 + 16 * h) = out;
+    }
 }
 
 /* ---- SHA-256 (one chain per lane), FIPS 180-4 -------------------------- */
@@ -679,9 +708,11 @@ hipError_t mec_launch_hash(int algo, const HashArgs *args,
         hipLaunchKernelGGL(sha256_batch_kernel, grid, blk, 0, stream, *args);
         break;
     case 2: /* HighwayHash256 */
-    case 3: /* HighwayHash256S: 2 lanes per chain (zipper pairs) */
-        grid.x = (uint32_t)((args->n_chains * 2 + 255) / 256);
-        hipLaunchKernelGGL(hh256_batch_kernel, grid, blk, 0, stream, *args);
+    case 3: /* HighwayHash256S: 2 lanes/chain (zipper pairs), 2 chains/lane
+               (ILP to cover the serial chain's dependency stalls) */
+        grid.x = (uint32_t)(((args->n_chains + 1) / 2 * 2 + 255) / 256);
+        hipLaunchKernelGGL(hh256_batch_kernel<2>, grid, blk, 0, stream,
+                           *args);
         break;
     case 4: /* BLAKE2b512 */
         hipLaunchKernelGGL(blake2b512_batch_kernel, grid, blk, 0, stream,
