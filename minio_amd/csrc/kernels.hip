@@ -342,24 +342,25 @@ __global__ void __launch_bounds__(256) hh256_batch_kernel(HashArgs a) {
         }
 
     int64_t len = a.msg_len;
-    /* 8-packet prefetch per chain, NC chains interleaved */
-    while (len >= 256) {
-        uint4 q[NC][8];
+    /* packet prefetch per chain, NC chains interleaved */
+    constexpr int DP = (NC == 1) ? 16 : 8; /* prefetch depth (packets) */
+    while (len >= 32 * DP) {
+        uint4 q[NC][DP];
 #pragma unroll
-        for (int t = 0; t < 8; t++)
+        for (int t = 0; t < DP; t++)
 #pragma unroll
             for (int u = 0; u < NC; u++)
                 q[u][t] = *(const uint4 *)(mp[u] + 32 * t);
 #pragma unroll
-        for (int t = 0; t < 8; t++)
+        for (int t = 0; t < DP; t++)
 #pragma unroll
             for (int u = 0; u < NC; u++)
                 hh2_update(s[u],
                            (uint64_t)q[u][t].x | ((uint64_t)q[u][t].y << 32),
                            (uint64_t)q[u][t].z | ((uint64_t)q[u][t].w << 32));
 #pragma unroll
-        for (int u = 0; u < NC; u++) mp[u] += 256;
-        len -= 256;
+        for (int u = 0; u < NC; u++) mp[u] += 32 * DP;
+        len -= 32 * DP;
     }
     while (len >= 32) {
 #pragma unroll
@@ -710,8 +711,8 @@ hipError_t mec_launch_hash(int algo, const HashArgs *args,
     case 2: /* HighwayHash256 */
     case 3: /* HighwayHash256S: 2 lanes/chain (zipper pairs), 2 chains/lane
                (ILP to cover the serial chain's dependency stalls) */
-        grid.x = (uint32_t)(((args->n_chains + 1) / 2 * 2 + 255) / 256);
-        hipLaunchKernelGGL(hh256_batch_kernel<2>, grid, blk, 0, stream,
+        grid.x = (uint32_t)((args->n_chains * 2 + 255) / 256);
+        hipLaunchKernelGGL(hh256_batch_kernel<1>, grid, blk, 0, stream,
                            *args);
         break;
     case 4: /* BLAKE2b512 */
