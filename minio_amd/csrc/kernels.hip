@@ -473,7 +473,9 @@ __device__ void sha256_block(uint32_t h[8], const uint32_t w_in[16]) {
     for (int i = 0; i < 16; i++) w[i] = w_in[i];
     uint32_t a = h[0], b = h[1], c = h[2], d = h[3], e = h[4], f = h[5],
              g = h[6], hh = h[7];
-#pragma unroll 4
+    /* full unroll: every w[] index must be compile-time, or the 16-word
+     * rolling schedule spills to scratch (measured 144 B/lane, 20x slower) */
+#pragma unroll
     for (int i = 0; i < 64; i++) {
         uint32_t wi;
         if (i < 16) {
@@ -498,53 +500,109 @@ __device__ void sha256_block(uint32_t h[8], const uint32_t w_in[16]) {
     h[4] += e; h[5] += f; h[6] += g; h[7] += hh;
 }
 
+/* NC independent chains per lane: SHA-256's round chain is strictly serial
+ * (measured ~17 cyc/instr effective at 1 chain/lane, 1 wave/SIMD — pure
+ * dependency latency); interleaving NC chains fills the stalls. */
+template <int NC>
 __global__ void __launch_bounds__(256) sha256_batch_kernel(HashArgs a) {
-    const int64_t chain = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (chain >= a.n_chains) return;
-    int64_t sum_idx;
-    const uint8_t *msg = chain_ptr(a, chain, sum_idx);
+    const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t c0 = tid * NC;
+    if (c0 >= a.n_chains) return;
 
-    uint32_t h[8] = {0x6a09e667, 0xbb67ae85, 0x3c6ef372, 0xa54ff53a,
-                     0x510e527f, 0x9b05688c, 0x1f83d9ab, 0x5be0cd19};
-    int64_t len = a.msg_len;
-    uint32_t w[16];
-    while (len >= 64) {
-        const uint4 *p = (const uint4 *)msg;
+    bool act[NC];
+    const uint8_t *mp[NC];
+    int64_t sum_idx[NC];
+    uint32_t h[NC][8];
 #pragma unroll
-        for (int q = 0; q < 4; q++) {
-            uint4 v = p[q];
-            w[4 * q + 0] = bswap32(v.x);
-            w[4 * q + 1] = bswap32(v.y);
-            w[4 * q + 2] = bswap32(v.z);
-            w[4 * q + 3] = bswap32(v.w);
+    for (int u = 0; u < NC; u++) {
+        act[u] = c0 + u < a.n_chains;
+        mp[u] = chain_ptr(a, act[u] ? c0 + u : c0, sum_idx[u]);
+        const uint32_t iv[8] = {0x6a09e667, 0xbb67ae85, 0x3c6ef372,
+                                0xa54ff53a, 0x510e527f, 0x9b05688c,
+                                0x1f83d9ab, 0x5be0cd19};
+#pragma unroll
+        for (int i = 0; i < 8; i++) h[u][i] = iv[i];
+    }
+    int64_t len = a.msg_len;
+    while (len >= 64) {
+        uint32_t w[NC][16];
+#pragma unroll
+        for (int u = 0; u < NC; u++) {
+            const uint4 *p = (const uint4 *)mp[u];
+#pragma unroll
+            for (int q = 0; q < 4; q++) {
+                uint4 v = p[q];
+                w[u][4 * q + 0] = bswap32(v.x);
+                w[u][4 * q + 1] = bswap32(v.y);
+                w[u][4 * q + 2] = bswap32(v.z);
+                w[u][4 * q + 3] = bswap32(v.w);
+            }
+            mp[u] += 64;
         }
-        sha256_block(h, w);
-        msg += 64;
+        uint32_t A[NC], B[NC], C[NC], D[NC], E[NC], F[NC], G[NC], H[NC];
+#pragma unroll
+        for (int u = 0; u < NC; u++) {
+            A[u] = h[u][0]; B[u] = h[u][1]; C[u] = h[u][2]; D[u] = h[u][3];
+            E[u] = h[u][4]; F[u] = h[u][5]; G[u] = h[u][6]; H[u] = h[u][7];
+        }
+#pragma unroll
+        for (int i = 0; i < 64; i++) {
+#pragma unroll
+            for (int u = 0; u < NC; u++) {
+                uint32_t wi;
+                if (i < 16) {
+                    wi = w[u][i];
+                } else {
+                    uint32_t w15 = w[u][(i - 15) & 15], w2 = w[u][(i - 2) & 15];
+                    uint32_t s0 = rotr32(w15, 7) ^ rotr32(w15, 18) ^ (w15 >> 3);
+                    uint32_t s1 = rotr32(w2, 17) ^ rotr32(w2, 19) ^ (w2 >> 10);
+                    wi = w[u][i & 15] + s0 + w[u][(i - 7) & 15] + s1;
+                    w[u][i & 15] = wi;
+                }
+                uint32_t S1 = rotr32(E[u], 6) ^ rotr32(E[u], 11) ^ rotr32(E[u], 25);
+                uint32_t ch = (E[u] & F[u]) ^ (~E[u] & G[u]);
+                uint32_t t1 = H[u] + S1 + ch + SHA_K[i] + wi;
+                uint32_t S0 = rotr32(A[u], 2) ^ rotr32(A[u], 13) ^ rotr32(A[u], 22);
+                uint32_t maj = (A[u] & B[u]) ^ (A[u] & C[u]) ^ (B[u] & C[u]);
+                uint32_t t2 = S0 + maj;
+                H[u] = G[u]; G[u] = F[u]; F[u] = E[u]; E[u] = D[u] + t1;
+                D[u] = C[u]; C[u] = B[u]; B[u] = A[u]; A[u] = t1 + t2;
+            }
+        }
+#pragma unroll
+        for (int u = 0; u < NC; u++) {
+            h[u][0] += A[u]; h[u][1] += B[u]; h[u][2] += C[u]; h[u][3] += D[u];
+            h[u][4] += E[u]; h[u][5] += F[u]; h[u][6] += G[u]; h[u][7] += H[u];
+        }
         len -= 64;
     }
     /* tail: rem bytes + 0x80 pad + 8-byte big-endian bit length */
-    {
+#pragma unroll
+    for (int u = 0; u < NC; u++) {
         uint8_t tail[128];
 #pragma unroll
         for (int i = 0; i < 128; i++) tail[i] = 0;
-        for (int i = 0; i < (int)len; i++) tail[i] = msg[i];
+        for (int i = 0; i < (int)len; i++) tail[i] = mp[u][i];
         tail[(int)len] = 0x80;
         const int tlen = (len < 56) ? 64 : 128;
         uint64_t bits = (uint64_t)a.msg_len * 8;
         for (int i = 0; i < 8; i++)
             tail[tlen - 1 - i] = (uint8_t)(bits >> (8 * i));
         for (int blk = 0; blk < tlen; blk += 64) {
+            uint32_t wt[16];
             for (int i = 0; i < 16; i++) {
                 const uint8_t *q = tail + blk + 4 * i;
-                w[i] = ((uint32_t)q[0] << 24) | ((uint32_t)q[1] << 16) |
-                       ((uint32_t)q[2] << 8) | q[3];
+                wt[i] = ((uint32_t)q[0] << 24) | ((uint32_t)q[1] << 16) |
+                        ((uint32_t)q[2] << 8) | q[3];
             }
-            sha256_block(h, w);
+            sha256_block(h[u], wt);
+        }
+        if (act[u]) {
+            uint8_t *out = a.sums + sum_idx[u] * 32;
+            for (int i = 0; i < 8; i++)
+                *(uint32_t *)(out + 4 * i) = bswap32(h[u][i]);
         }
     }
-    uint8_t *out = a.sums + sum_idx * 32;
-    for (int i = 0; i < 8; i++)
-        *(uint32_t *)(out + 4 * i) = bswap32(h[i]);
 }
 
 /* ---- BLAKE2b-512 (one chain per lane), RFC 7693 ------------------------ */
@@ -705,8 +763,11 @@ hipError_t mec_launch_hash(int algo, const HashArgs *args,
     dim3 grid((uint32_t)blocks);
     dim3 blk(256);
     switch (algo) {
-    case 1: /* SHA256 */
-        hipLaunchKernelGGL(sha256_batch_kernel, grid, blk, 0, stream, *args);
+    case 1: /* SHA256: 2 chains per lane (serial rounds -> ILP; NC=4
+               spills to scratch at 217 VGPR) */
+        grid.x = (uint32_t)(((args->n_chains + 1) / 2 + 255) / 256);
+        hipLaunchKernelGGL(sha256_batch_kernel<2>, grid, blk, 0, stream,
+                           *args);
         break;
     case 2: /* HighwayHash256 */
     case 3: /* HighwayHash256S: 2 lanes/chain (zipper pairs), 2 chains/lane
