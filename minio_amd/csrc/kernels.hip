@@ -57,19 +57,25 @@ __device__ __forceinline__ void xor4(uint4 &a, const uint4 &b) {
 template <int D, int P, const uint8_t (&MAT)[P][D]>
 __global__ void __launch_bounds__(256) gf_encode_kernel(GfEncArgs a) {
     const int b = blockIdx.y;
-    const int64_t cols = (a.shard_len + 15) >> 4;
+    /* 32 B per thread (two uint4 columns share one doubling ladder pass) */
+    const int64_t cols = (a.shard_len + 31) >> 5;
     const uint8_t *__restrict__ sbase = a.data + (int64_t)b * D * a.row_stride;
     uint8_t *__restrict__ obase = a.parity + (int64_t)b * P * a.row_stride;
 
     for (int64_t c = blockIdx.x * blockDim.x + threadIdx.x; c < cols;
          c += (int64_t)gridDim.x * blockDim.x) {
-        const int64_t j = c << 4;
-        uint4 acc[P];
+        const int64_t j = c << 5;
+        uint4 acc0[P], acc1[P];
 #pragma unroll
-        for (int i = 0; i < P; i++) acc[i] = uint4{0, 0, 0, 0};
+        for (int i = 0; i < P; i++) {
+            acc0[i] = uint4{0, 0, 0, 0};
+            acc1[i] = uint4{0, 0, 0, 0};
+        }
 #pragma unroll
         for (int k = 0; k < D; k++) {
-            uint4 pw = *(const uint4 *)(sbase + (int64_t)k * a.row_stride + j);
+            const uint8_t *row = sbase + (int64_t)k * a.row_stride + j;
+            uint4 pw0 = *(const uint4 *)row;
+            uint4 pw1 = *(const uint4 *)(row + 16);
 #pragma unroll
             for (int bit = 0; bit < 8; bit++) {
                 uint32_t need = 0;
@@ -77,15 +83,24 @@ __global__ void __launch_bounds__(256) gf_encode_kernel(GfEncArgs a) {
                 for (int i = 0; i < P; i++)
                     need |= (uint32_t)MAT[i][k] >> bit;
                 if (!need) break; /* compile-time folded */
-                if (bit) gf2x4(pw);
+                if (bit) {
+                    gf2x4(pw0);
+                    gf2x4(pw1);
+                }
 #pragma unroll
                 for (int i = 0; i < P; i++)
-                    if ((MAT[i][k] >> bit) & 1) xor4(acc[i], pw);
+                    if ((MAT[i][k] >> bit) & 1) {
+                        xor4(acc0[i], pw0);
+                        xor4(acc1[i], pw1);
+                    }
             }
         }
 #pragma unroll
-        for (int i = 0; i < P; i++)
-            *(uint4 *)(obase + (int64_t)i * a.row_stride + j) = acc[i];
+        for (int i = 0; i < P; i++) {
+            uint8_t *orow = obase + (int64_t)i * a.row_stride + j;
+            *(uint4 *)orow = acc0[i];
+            *(uint4 *)(orow + 16) = acc1[i];
+        }
     }
 }
 
@@ -342,26 +357,48 @@ __global__ void __launch_bounds__(256) hh256_batch_kernel(HashArgs a) {
         }
 
     int64_t len = a.msg_len;
-    /* packet prefetch per chain, NC chains interleaved */
+    /* double-buffered packet prefetch: compute tile A while tile B's loads
+     * are in flight (a single-buffer loop leaves ~36% of wave-cycles in
+     * memory waits — r03 PMC).  Buffers alternate by explicit two-phase
+     * unroll: a runtime-selected array reference would spill to scratch. */
     constexpr int DP = (NC == 1) ? 16 : 8; /* prefetch depth (packets) */
-    while (len >= 32 * DP) {
-        uint4 q[NC][DP];
-#pragma unroll
-        for (int t = 0; t < DP; t++)
-#pragma unroll
-            for (int u = 0; u < NC; u++)
-                q[u][t] = *(const uint4 *)(mp[u] + 32 * t);
-#pragma unroll
-        for (int t = 0; t < DP; t++)
-#pragma unroll
-            for (int u = 0; u < NC; u++)
-                hh2_update(s[u],
-                           (uint64_t)q[u][t].x | ((uint64_t)q[u][t].y << 32),
-                           (uint64_t)q[u][t].z | ((uint64_t)q[u][t].w << 32));
-#pragma unroll
-        for (int u = 0; u < NC; u++) mp[u] += 32 * DP;
-        len -= 32 * DP;
+    uint4 qa[NC][DP], qb[NC][DP];
+#define HH_LOAD(Q)                                                           \
+    {                                                                        \
+        _Pragma("unroll") for (int t = 0; t < DP; t++)                       \
+            _Pragma("unroll") for (int u = 0; u < NC; u++)                   \
+                Q[u][t] = *(const uint4 *)(mp[u] + 32 * t);                  \
+        _Pragma("unroll") for (int u = 0; u < NC; u++) mp[u] += 32 * DP;     \
     }
+#define HH_COMP(Q)                                                           \
+    {                                                                        \
+        _Pragma("unroll") for (int t = 0; t < DP; t++)                       \
+            _Pragma("unroll") for (int u = 0; u < NC; u++)                   \
+                hh2_update(s[u],                                             \
+                           (uint64_t)Q[u][t].x | ((uint64_t)Q[u][t].y << 32),\
+                           (uint64_t)Q[u][t].z | ((uint64_t)Q[u][t].w << 32));\
+    }
+    if (len >= 32 * DP) {
+        HH_LOAD(qa)
+        len -= 32 * DP;
+        while (len >= 2 * 32 * DP) {
+            HH_LOAD(qb)
+            HH_COMP(qa)
+            HH_LOAD(qa)
+            HH_COMP(qb)
+            len -= 2 * 32 * DP;
+        }
+        if (len >= 32 * DP) {
+            HH_LOAD(qb)
+            HH_COMP(qa)
+            HH_COMP(qb)
+            len -= 32 * DP;
+        } else {
+            HH_COMP(qa)
+        }
+    }
+#undef HH_LOAD
+#undef HH_COMP
     while (len >= 32) {
 #pragma unroll
         for (int u = 0; u < NC; u++) {
@@ -715,7 +752,7 @@ extern "C" {
  * compiled specialization (caller falls back to the generic kernel). */
 hipError_t mec_launch_gf_encode_spec(int d, int p, const GfEncArgs *args,
                                      int n, hipStream_t stream) {
-    const int64_t cols = (args->shard_len + 15) >> 4;
+    const int64_t cols = (args->shard_len + 31) >> 5;
     int64_t max_x = (cols + 255) / 256;
     int64_t want_x = (2048 + n - 1) / n;
     int64_t blocks_x = want_x < max_x ? want_x : max_x;
