@@ -226,14 +226,11 @@ def main():
         else:
             t_gf = timed(lambda: check(lib.mec_encode_batch_dev_async(
                 ctx, n, dev_data, bs, dev_par, algo, None)))
-            # hash leg measured directly (the fused call overlaps hash(data)
-            # with the GF kernel, so fused-minus-gf would under-report)
-            def hash_all():
-                check(lib.mec_bitrot_sum_batch_dev(
-                    ctx, algo, n * d, dev_data, S, stride, dev_sum))
-                check(lib.mec_bitrot_sum_batch_dev(
-                    ctx, algo, n * p, dev_par, S, stride, dev_sum))
-            t_hash = timed(hash_all)
+            t_fused = timed(lambda: check(lib.mec_encode_batch_dev_async(
+                ctx, n, dev_data, bs, dev_par, algo, dev_sum)))
+            # kernels run sequentially on one stream, so the hash leg is
+            # the fused-minus-gf difference
+            t_hash = max(t_fused - t_gf, 1e-9)
             # per-launch algorithmic bytes:
             gf_bytes = n * (bs + p * S)          # read data, write parity
             hash_bytes = n * (total * S)         # read every shard once
@@ -249,6 +246,9 @@ def main():
             "unit": "GB/s", "frac": round(achieved / peak, 4),
             "traffic": None,
         }
+        if not is_decode:
+            roofline["legs_ms"] = {"gf": round(t_gf * 1e3, 3),
+                                   "hash": round(t_hash * 1e3, 3)}
 
     # ---- CPU baseline: oracle timed on host cores (bounded sample) ----
     cpu_baseline = None
