@@ -20,6 +20,7 @@
  */
 #include <hip/hip_runtime.h>
 #include <cstdint>
+#include <cstdlib>
 
 #include "kernels.h"
 
@@ -54,28 +55,31 @@ __device__ __forceinline__ void xor4(uint4 &a, const uint4 &b) {
  * doubling steps stop at each column's highest set bit. */
 #include "ec_matrices_gen.h"
 
-template <int D, int P, const uint8_t (&MAT)[P][D]>
+/* W = uint4 columns per thread (16 or 32 B); NT = nontemporal parity
+ * stores (parity is written once, never re-read by this kernel).
+ * Variant selection via MEC_GF_W / MEC_GF_NT env (perf sweeps). */
+template <int D, int P, const uint8_t (&MAT)[P][D], int W, bool NT>
 __global__ void __launch_bounds__(256) gf_encode_kernel(GfEncArgs a) {
     const int b = blockIdx.y;
-    /* 32 B per thread (two uint4 columns share one doubling ladder pass) */
-    const int64_t cols = (a.shard_len + 31) >> 5;
+    const int64_t cols = (a.shard_len + 16 * W - 1) / (16 * W);
     const uint8_t *__restrict__ sbase = a.data + (int64_t)b * D * a.row_stride;
     uint8_t *__restrict__ obase = a.parity + (int64_t)b * P * a.row_stride;
 
     for (int64_t c = blockIdx.x * blockDim.x + threadIdx.x; c < cols;
          c += (int64_t)gridDim.x * blockDim.x) {
-        const int64_t j = c << 5;
-        uint4 acc0[P], acc1[P];
+        const int64_t j = c * (16 * W);
+        uint4 acc[P][W];
 #pragma unroll
-        for (int i = 0; i < P; i++) {
-            acc0[i] = uint4{0, 0, 0, 0};
-            acc1[i] = uint4{0, 0, 0, 0};
-        }
+        for (int i = 0; i < P; i++)
+#pragma unroll
+            for (int w = 0; w < W; w++) acc[i][w] = uint4{0, 0, 0, 0};
 #pragma unroll
         for (int k = 0; k < D; k++) {
             const uint8_t *row = sbase + (int64_t)k * a.row_stride + j;
-            uint4 pw0 = *(const uint4 *)row;
-            uint4 pw1 = *(const uint4 *)(row + 16);
+            uint4 pw[W];
+#pragma unroll
+            for (int w = 0; w < W; w++)
+                pw[w] = *(const uint4 *)(row + 16 * w);
 #pragma unroll
             for (int bit = 0; bit < 8; bit++) {
                 uint32_t need = 0;
@@ -83,23 +87,30 @@ __global__ void __launch_bounds__(256) gf_encode_kernel(GfEncArgs a) {
                 for (int i = 0; i < P; i++)
                     need |= (uint32_t)MAT[i][k] >> bit;
                 if (!need) break; /* compile-time folded */
-                if (bit) {
-                    gf2x4(pw0);
-                    gf2x4(pw1);
-                }
+                if (bit)
+#pragma unroll
+                    for (int w = 0; w < W; w++) gf2x4(pw[w]);
 #pragma unroll
                 for (int i = 0; i < P; i++)
-                    if ((MAT[i][k] >> bit) & 1) {
-                        xor4(acc0[i], pw0);
-                        xor4(acc1[i], pw1);
-                    }
+                    if ((MAT[i][k] >> bit) & 1)
+#pragma unroll
+                        for (int w = 0; w < W; w++) xor4(acc[i][w], pw[w]);
             }
         }
 #pragma unroll
         for (int i = 0; i < P; i++) {
             uint8_t *orow = obase + (int64_t)i * a.row_stride + j;
-            *(uint4 *)orow = acc0[i];
-            *(uint4 *)(orow + 16) = acc1[i];
+#pragma unroll
+            for (int w = 0; w < W; w++) {
+                if (NT) {
+                    typedef unsigned int v4u __attribute__((ext_vector_type(4)));
+                    v4u v = {acc[i][w].x, acc[i][w].y, acc[i][w].z,
+                             acc[i][w].w};
+                    __builtin_nontemporal_store(v, (v4u *)(orow + 16 * w));
+                } else {
+                    *(uint4 *)(orow + 16 * w) = acc[i][w];
+                }
+            }
         }
     }
 }
@@ -750,19 +761,42 @@ extern "C" {
 
 /* Specialized encode launch; returns hipErrorNotSupported when (d,p) has no
  * compiled specialization (caller falls back to the generic kernel). */
+static int gf_env_int(const char *name, int dflt) {
+    const char *v = getenv(name);
+    return v ? atoi(v) : dflt;
+}
+
 hipError_t mec_launch_gf_encode_spec(int d, int p, const GfEncArgs *args,
                                      int n, hipStream_t stream) {
-    const int64_t cols = (args->shard_len + 31) >> 5;
+    static const int env_w = gf_env_int("MEC_GF_W", 2);
+    static const int env_nt = gf_env_int("MEC_GF_NT", 0);
+    static const int env_wgx = gf_env_int("MEC_GF_WGX", 1);
+    const int W = (env_w == 1) ? 1 : 2;
+    const int64_t cols = (args->shard_len + 16 * W - 1) / (16 * W);
     int64_t max_x = (cols + 255) / 256;
-    int64_t want_x = (2048 + n - 1) / n;
+    int64_t want_x = ((int64_t)2048 * env_wgx + n - 1) / n;
     int64_t blocks_x = want_x < max_x ? want_x : max_x;
     if (blocks_x < 1) blocks_x = 1;
     dim3 grid((uint32_t)blocks_x, n);
     dim3 blk(256);
 #define X(D, P)                                                              \
     if (d == D && p == P) {                                                  \
-        hipLaunchKernelGGL((gf_encode_kernel<D, P, MAT_##D##_##P>), grid,    \
-                           blk, 0, stream, *args);                           \
+        if (W == 1 && !env_nt)                                               \
+            hipLaunchKernelGGL((gf_encode_kernel<D, P, MAT_##D##_##P, 1,     \
+                                                 false>),                    \
+                               grid, blk, 0, stream, *args);                 \
+        else if (W == 1)                                                     \
+            hipLaunchKernelGGL((gf_encode_kernel<D, P, MAT_##D##_##P, 1,     \
+                                                 true>),                     \
+                               grid, blk, 0, stream, *args);                 \
+        else if (!env_nt)                                                    \
+            hipLaunchKernelGGL((gf_encode_kernel<D, P, MAT_##D##_##P, 2,     \
+                                                 false>),                    \
+                               grid, blk, 0, stream, *args);                 \
+        else                                                                 \
+            hipLaunchKernelGGL((gf_encode_kernel<D, P, MAT_##D##_##P, 2,     \
+                                                 true>),                     \
+                               grid, blk, 0, stream, *args);                 \
         return hipGetLastError();                                            \
     }
     MEC_SPECIALIZED_GEOS(X)
