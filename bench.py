@@ -87,7 +87,10 @@ def main():
     hsz = 32
     lib = minio_amd._lib
 
-    e = minio_amd.Erasure(d, p, bs, device=local_rank)
+    # local_rank maps 1:1 onto GPUs on a real node; the modulo only matters
+    # when testing multi-rank on a smaller box
+    dev = local_rank % max(1, minio_amd.device_count())
+    e = minio_amd.Erasure(d, p, bs, device=dev)
     ctx = e._ctx
 
     # ---- stage synthetic inputs into HBM (outside the timed region) ----

@@ -768,9 +768,9 @@ static int gf_env_int(const char *name, int dflt) {
 
 hipError_t mec_launch_gf_encode_spec(int d, int p, const GfEncArgs *args,
                                      int n, hipStream_t stream) {
-    static const int env_w = gf_env_int("MEC_GF_W", 2);
-    static const int env_nt = gf_env_int("MEC_GF_NT", 0);
-    static const int env_wgx = gf_env_int("MEC_GF_WGX", 1);
+    static const int env_w = gf_env_int("MEC_GF_W", 1);
+    static const int env_nt = gf_env_int("MEC_GF_NT", 1);
+    static const int env_wgx = gf_env_int("MEC_GF_WGX", 4);
     const int W = (env_w == 1) ? 1 : 2;
     const int64_t cols = (args->shard_len + 16 * W - 1) / (16 * W);
     int64_t max_x = (cols + 255) / 256;
