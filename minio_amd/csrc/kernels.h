@@ -52,7 +52,20 @@ struct GfEncArgs {
     int64_t shard_len;
 };
 
+/* Fused single-pass encode+HighwayHash (fused.hip) */
+struct FusedArgs {
+    const uint8_t *data; /* n * d * row_stride */
+    uint8_t *parity;     /* n * p * row_stride (output) */
+    uint8_t *sums;       /* n * (d+p) * 32 (output) */
+    int64_t row_stride;
+    int64_t shard_len;
+    int64_t n;
+    uint64_t key[4];
+};
+
 extern "C" {
+hipError_t mec_launch_fused_encode_hh(int d, int p, const FusedArgs *args,
+                                      hipStream_t stream);
 hipError_t mec_launch_gf_matmul(const GfMatmulArgs *args, int n_dst, int n,
                                 hipStream_t stream);
 hipError_t mec_launch_gf_encode_spec(int d, int p, const GfEncArgs *args,
