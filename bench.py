@@ -105,9 +105,12 @@ def main():
             raise RuntimeError(f"mec status {st}: {lib.mec_last_error()}")
 
     dev_data, dev_par, dev_sum = vp(), vp(), vp()
+    dev_par2, dev_sum2 = vp(), vp()
     check(lib.mec_dev_alloc(ctx, data_bytes, ctypes.byref(dev_data)))
     check(lib.mec_dev_alloc(ctx, par_bytes, ctypes.byref(dev_par)))
     check(lib.mec_dev_alloc(ctx, sum_bytes, ctypes.byref(dev_sum)))
+    check(lib.mec_dev_alloc(ctx, par_bytes, ctypes.byref(dev_par2)))
+    check(lib.mec_dev_alloc(ctx, sum_bytes, ctypes.byref(dev_sum2)))
 
     # host-generate, scatter into padded strided layout, one H2D copy
     chunk_blocks = max(1, (256 << 20) // (d * stride))
@@ -164,10 +167,23 @@ def main():
                 ctx, vp(dev_shards.value + b * total * stride), 0,
                 n_er * stride))
 
+    # pipelined encode: alternate parity/sums buffer sets per step so batch
+    # t's hash overlaps batch t+1's GF (see mec_encode_batch_dev_pipe
+    # contract); each step still performs the full fused work for its batch
+    use_pipe = (not is_decode) and algo in (2, 3) \
+        and os.environ.get("MEC_PIPE", "1") != "0"
+    step_no = [0]
+
     def step():
         if is_decode:
             check(lib.mec_reconstruct_batch_dev_async(
                 ctx, n, dev_shards, present, S, 1))
+        elif use_pipe:
+            par = dev_par if step_no[0] % 2 == 0 else dev_par2
+            sm = dev_sum if step_no[0] % 2 == 0 else dev_sum2
+            step_no[0] += 1
+            check(lib.mec_encode_batch_dev_pipe(
+                ctx, n, dev_data, bs, par, algo, sm))
         else:
             check(lib.mec_encode_batch_dev_async(
                 ctx, n, dev_data, bs, dev_par, algo, dev_sum))
@@ -175,7 +191,7 @@ def main():
     # ---- warmup ----
     for _ in range(args.warmup):
         step()
-    check(lib.mec_stream_sync(ctx))
+    check(lib.mec_pipe_sync(ctx) if use_pipe else lib.mec_stream_sync(ctx))
 
     # ---- timed region: barrier + sync both sides, max over ranks ----
     if dist:
@@ -185,6 +201,8 @@ def main():
     t0 = time.perf_counter()
     for _ in range(args.steps):
         step()
+    if use_pipe:
+        check(lib.mec_pipe_sync(ctx))
     ms = ctypes.c_float()
     check(lib.mec_timer_stop(ctx, ctypes.byref(ms)))
     check(lib.mec_stream_sync(ctx))

@@ -125,6 +125,19 @@ mec_status mec_encode_batch_dev_async(mec_ctx *ctx, int n,
                                       void *parity_dev, int bitrot_algo,
                                       void *sums_dev);
 
+/* Pipelined encode: batch t's hash overlaps batch t+1's GF kernel (the
+ * hash is latency-bound per chain and leaves most SIMDs idle; independent
+ * batches — i.e. different objects — fill them).  CONTRACT: alternate two
+ * parity/sums buffer sets in strict round-robin and call mec_pipe_sync
+ * before reading results.  Per-call results are identical to
+ * mec_encode_batch_dev.  Requires a compiled (d,p) specialization and a
+ * HighwayHash algorithm. */
+mec_status mec_encode_batch_dev_pipe(mec_ctx *ctx, int n,
+                                     const void *data_dev, int64_t block_len,
+                                     void *parity_dev, int bitrot_algo,
+                                     void *sums_dev);
+mec_status mec_pipe_sync(mec_ctx *ctx);
+
 /* ---- batch reconstruct (ReconstructData / Reconstruct / Heal kernel) ---
  *
  * shards_dev: n * (d+p) * stride bytes; present[i] != 0 marks shard row i

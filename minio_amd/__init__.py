@@ -92,6 +92,8 @@ def _load():
     lib.mec_encode_batch.argtypes = [vp, i32, u8p, i64, u8p, i32, u8p]
     lib.mec_encode_batch_dev.argtypes = [vp, i32, vp, i64, vp, i32, vp]
     lib.mec_encode_batch_dev_async.argtypes = [vp, i32, vp, i64, vp, i32, vp]
+    lib.mec_encode_batch_dev_pipe.argtypes = [vp, i32, vp, i64, vp, i32, vp]
+    lib.mec_pipe_sync.argtypes = [vp]
     lib.mec_reconstruct_batch.argtypes = [vp, i32, u8p, u8p, i64, i32]
     lib.mec_reconstruct_batch_dev.argtypes = [vp, i32, vp, u8p, i64, i32]
     lib.mec_reconstruct_batch_dev_async.argtypes = [vp, i32, vp, u8p, i64, i32]

@@ -78,6 +78,8 @@ struct mec_ctx {
     hipStream_t stream2 = nullptr; /* hash(data) overlap lane */
     hipEvent_t ev_start = nullptr, ev_stop = nullptr;
     hipEvent_t ev_gf = nullptr, ev_h = nullptr, ev_fork = nullptr;
+    hipEvent_t ev_pipe[2] = {nullptr, nullptr}; /* hash-done per buffer slot */
+    int pipe_idx = 0;
     std::mutex mu;
 
     /* grow-only scratch (device + pinned host) for host-pointer calls */
@@ -194,6 +196,8 @@ mec_status mec_ctx_create(int d, int p, int64_t block_size, int device,
     if (e == hipSuccess) e = hipEventCreateWithFlags(&ctx->ev_gf, hipEventDisableTiming);
     if (e == hipSuccess) e = hipEventCreateWithFlags(&ctx->ev_h, hipEventDisableTiming);
     if (e == hipSuccess) e = hipEventCreateWithFlags(&ctx->ev_fork, hipEventDisableTiming);
+    if (e == hipSuccess) e = hipEventCreateWithFlags(&ctx->ev_pipe[0], hipEventDisableTiming);
+    if (e == hipSuccess) e = hipEventCreateWithFlags(&ctx->ev_pipe[1], hipEventDisableTiming);
     if (e != hipSuccess) {
         set_err("ctx_create", e);
         delete ctx;
@@ -219,6 +223,8 @@ void mec_ctx_destroy(mec_ctx *ctx) {
     if (ctx->ev_gf) (void)hipEventDestroy(ctx->ev_gf);
     if (ctx->ev_h) (void)hipEventDestroy(ctx->ev_h);
     if (ctx->ev_fork) (void)hipEventDestroy(ctx->ev_fork);
+    if (ctx->ev_pipe[0]) (void)hipEventDestroy(ctx->ev_pipe[0]);
+    if (ctx->ev_pipe[1]) (void)hipEventDestroy(ctx->ev_pipe[1]);
     if (ctx->stream2) (void)hipStreamDestroy(ctx->stream2);
     if (ctx->stream) (void)hipStreamDestroy(ctx->stream);
     delete ctx;
@@ -312,6 +318,70 @@ gf_done:
         memcpy(h.key, kMagicHHKey, 32);
         HIP_TRY(mec_launch_hash(algo, &h, ctx->stream));
     }
+    return MEC_OK;
+}
+
+/* Pipelined encode: batch t's hash (on the context's second stream)
+ * overlaps batch t+1's GF.  CONTRACT: the caller alternates between TWO
+ * parity/sums buffer sets in strict round-robin and calls mec_pipe_sync
+ * before reading any results.  Results are identical to
+ * mec_encode_batch_dev per call; only cross-call scheduling differs
+ * (independent batches — MinIO objects — pipeline the same way). */
+mec_status mec_encode_batch_dev_pipe(mec_ctx *ctx, int n,
+                                     const void *data_dev, int64_t block_len,
+                                     void *parity_dev, int algo,
+                                     void *sums_dev) {
+    std::lock_guard<std::mutex> lk(ctx->mu);
+    if (n <= 0 || block_len <= 0 || block_len > ctx->block_size ||
+        sums_dev == nullptr || !hash_size(algo))
+        return MEC_ERR_INVALID_ARG;
+    HIP_TRY(hipSetDevice(ctx->device));
+    const int d = ctx->d, p = ctx->p;
+    const int64_t S_call = ceil_frac(block_len, d);
+    const int idx = ctx->pipe_idx & 1;
+    ctx->pipe_idx ^= 1;
+    /* this buffer slot's previous hash must be drained before GF rewrites
+     * the parity buffer */
+    HIP_TRY(hipStreamWaitEvent(ctx->stream, ctx->ev_pipe[idx], 0));
+    {
+        GfEncArgs ea{};
+        ea.data = (const uint8_t *)data_dev;
+        ea.parity = (uint8_t *)parity_dev;
+        ea.row_stride = ctx->stride;
+        ea.shard_len = S_call;
+        hipError_t he = mec_launch_gf_encode_spec(d, p, &ea, n, ctx->stream);
+        if (he == hipErrorNotSupported)
+            return MEC_ERR_INVALID_ARG; /* pipe mode needs a specialization */
+        if (he != hipSuccess) {
+            set_err("gf_encode_spec", he);
+            return MEC_ERR_HIP;
+        }
+    }
+    HIP_TRY(hipEventRecord(ctx->ev_gf, ctx->stream));
+    HIP_TRY(hipStreamWaitEvent(ctx->stream2, ctx->ev_gf, 0));
+    {
+        HashArgs h{};
+        h.data = (const uint8_t *)data_dev;
+        h.parity = (const uint8_t *)parity_dev;
+        h.sums = (uint8_t *)sums_dev;
+        h.row_stride = ctx->stride;
+        h.msg_len = S_call;
+        h.d = d;
+        h.p = p;
+        h.mode = MEC_HASH_ALL;
+        h.n_chains = (int64_t)n * (d + p);
+        memcpy(h.key, kMagicHHKey, 32);
+        HIP_TRY(mec_launch_hash(algo, &h, ctx->stream2));
+    }
+    HIP_TRY(hipEventRecord(ctx->ev_pipe[idx], ctx->stream2));
+    return MEC_OK;
+}
+
+mec_status mec_pipe_sync(mec_ctx *ctx) {
+    std::lock_guard<std::mutex> lk(ctx->mu);
+    HIP_TRY(hipSetDevice(ctx->device));
+    HIP_TRY(hipStreamSynchronize(ctx->stream));
+    HIP_TRY(hipStreamSynchronize(ctx->stream2));
     return MEC_OK;
 }
 

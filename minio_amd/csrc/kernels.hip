@@ -389,6 +389,9 @@ __global__ void __launch_bounds__(256) hh256_batch_kernel(HashArgs a) {
                            (uint64_t)Q[u][t].x | ((uint64_t)Q[u][t].y << 32),\
                            (uint64_t)Q[u][t].z | ((uint64_t)Q[u][t].w << 32));\
     }
+    /* raise wave priority: when a GF kernel shares the CU (pipelined
+     * steps), the latency-critical chains must win issue slots (T5) */
+    __builtin_amdgcn_s_setprio(1);
     if (len >= 32 * DP) {
         HH_LOAD(qa)
         len -= 32 * DP;
@@ -410,6 +413,7 @@ __global__ void __launch_bounds__(256) hh256_batch_kernel(HashArgs a) {
     }
 #undef HH_LOAD
 #undef HH_COMP
+    __builtin_amdgcn_s_setprio(0);
     while (len >= 32) {
 #pragma unroll
         for (int u = 0; u < NC; u++) {
