@@ -775,6 +775,10 @@ hipError_t mec_launch_gf_encode_spec(int d, int p, const GfEncArgs *args,
     static const int env_w = gf_env_int("MEC_GF_W", 1);
     static const int env_nt = gf_env_int("MEC_GF_NT", 1);
     static const int env_wgx = gf_env_int("MEC_GF_WGX", 4);
+    /* dummy dynamic-LDS reservation caps GF occupancy per CU so the
+     * latency-critical hash waves of the pipelined previous batch keep
+     * issue slots (0 = uncapped) */
+    static const int env_lds = gf_env_int("MEC_GF_LDS", 0);
     const int W = (env_w == 1) ? 1 : 2;
     const int64_t cols = (args->shard_len + 16 * W - 1) / (16 * W);
     int64_t max_x = (cols + 255) / 256;
@@ -788,19 +792,19 @@ hipError_t mec_launch_gf_encode_spec(int d, int p, const GfEncArgs *args,
         if (W == 1 && !env_nt)                                               \
             hipLaunchKernelGGL((gf_encode_kernel<D, P, MAT_##D##_##P, 1,     \
                                                  false>),                    \
-                               grid, blk, 0, stream, *args);                 \
+                               grid, blk, env_lds, stream, *args);           \
         else if (W == 1)                                                     \
             hipLaunchKernelGGL((gf_encode_kernel<D, P, MAT_##D##_##P, 1,     \
                                                  true>),                     \
-                               grid, blk, 0, stream, *args);                 \
+                               grid, blk, env_lds, stream, *args);           \
         else if (!env_nt)                                                    \
             hipLaunchKernelGGL((gf_encode_kernel<D, P, MAT_##D##_##P, 2,     \
                                                  false>),                    \
-                               grid, blk, 0, stream, *args);                 \
+                               grid, blk, env_lds, stream, *args);           \
         else                                                                 \
             hipLaunchKernelGGL((gf_encode_kernel<D, P, MAT_##D##_##P, 2,     \
                                                  true>),                     \
-                               grid, blk, 0, stream, *args);                 \
+                               grid, blk, env_lds, stream, *args);           \
         return hipGetLastError();                                            \
     }
     MEC_SPECIALIZED_GEOS(X)
