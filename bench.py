@@ -170,7 +170,7 @@ def main():
     # pipelined encode: alternate parity/sums buffer sets per step so batch
     # t's hash overlaps batch t+1's GF (see mec_encode_batch_dev_pipe
     # contract); each step still performs the full fused work for its batch
-    use_pipe = (not is_decode) and algo in (2, 3) \
+    use_pipe = (not is_decode) and algo in (1, 2, 3) \
         and os.environ.get("MEC_PIPE", "1") != "0"
     step_no = [0]
 
