@@ -170,7 +170,10 @@ def main():
     # pipelined encode: alternate parity/sums buffer sets per step so batch
     # t's hash overlaps batch t+1's GF (see mec_encode_batch_dev_pipe
     # contract); each step still performs the full fused work for its batch
-    use_pipe = (not is_decode) and algo in (1, 2, 3) \
+    # HighwayHash only: SHA-256's chains are so latency-bound that GF
+    # co-residency slows them more than the overlap saves (13.4 vs 10.9
+    # ms/step measured)
+    use_pipe = (not is_decode) and algo in (2, 3) \
         and os.environ.get("MEC_PIPE", "1") != "0"
     step_no = [0]
 
