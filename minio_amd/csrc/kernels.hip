@@ -334,7 +334,7 @@ __device__ __forceinline__ void hh2_update(HH2 &s, uint64_t w0, uint64_t w1) {
  * single chain per lane leaves every dependent-op latency exposed (measured
  * ~340 cyc/packet vs ~104 issue-bound).  Interleaving NC independent
  * chains' packet updates in one lane fills those stalls. */
-template <int NC>
+template <int NC, bool RAGGED>
 __global__ void __launch_bounds__(256) hh256_batch_kernel(HashArgs a) {
     const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     const int64_t slot = tid >> 1;
@@ -424,8 +424,11 @@ __global__ void __launch_bounds__(256) hh256_batch_kernel(HashArgs a) {
         }
         len -= 32;
     }
-    if (len > 0) {
-        /* UpdateRemainder (published portable semantics); tail-only cost */
+    if (RAGGED && len > 0) {
+        /* UpdateRemainder (published portable semantics); tail-only cost.
+         * Compiled out for 32-aligned launches: the byte-wise packet
+         * builder costs ~900 B of SGPR spill when duplicated per chain
+         * (the NC=2 regression), so aligned lengths get the lean kernel. */
         const int mod32 = (int)len;
         const int mod4 = mod32 & 3;
 #pragma unroll
@@ -849,11 +852,19 @@ hipError_t mec_launch_hash(int algo, const HashArgs *args,
                            *args);
         break;
     case 2: /* HighwayHash256 */
-    case 3: /* HighwayHash256S: 2 lanes/chain (zipper pairs), 2 chains/lane
-               (ILP to cover the serial chain's dependency stalls) */
-        grid.x = (uint32_t)((args->n_chains * 2 + 255) / 256);
-        hipLaunchKernelGGL(hh256_batch_kernel<1>, grid, blk, 0, stream,
-                           *args);
+    case 3: /* HighwayHash256S: 2 lanes/chain (zipper pairs).  32-aligned
+               lengths run 2 chains/lane (ILP over the serial chain's
+               dependency stalls) with the ragged-tail code compiled out;
+               ragged lengths take the 1-chain/lane full kernel. */
+        if (args->msg_len % 32 == 0) {
+            grid.x = (uint32_t)(((args->n_chains + 1) / 2 * 2 + 255) / 256);
+            hipLaunchKernelGGL((hh256_batch_kernel<2, false>), grid, blk, 0,
+                               stream, *args);
+        } else {
+            grid.x = (uint32_t)((args->n_chains * 2 + 255) / 256);
+            hipLaunchKernelGGL((hh256_batch_kernel<1, true>), grid, blk, 0,
+                               stream, *args);
+        }
         break;
     case 4: /* BLAKE2b512 */
         hipLaunchKernelGGL(blake2b512_batch_kernel, grid, blk, 0, stream,
