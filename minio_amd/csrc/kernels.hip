@@ -856,9 +856,13 @@ hipError_t mec_launch_hash(int algo, const HashArgs *args,
                lengths run 2 chains/lane (ILP over the serial chain's
                dependency stalls) with the ragged-tail code compiled out;
                ragged lengths take the 1-chain/lane full kernel. */
+        /* NC=2 measured 2.3x slower even with the tail compiled out (the
+         * doubled stream reaches 256 VGPR and the interleave does not
+         * cover HH's dependency chains the way it does SHA's) — NC=1 for
+         * both cases; aligned lengths still skip the tail code. */
         if (args->msg_len % 32 == 0) {
-            grid.x = (uint32_t)(((args->n_chains + 1) / 2 * 2 + 255) / 256);
-            hipLaunchKernelGGL((hh256_batch_kernel<2, false>), grid, blk, 0,
+            grid.x = (uint32_t)((args->n_chains * 2 + 255) / 256);
+            hipLaunchKernelGGL((hh256_batch_kernel<1, false>), grid, blk, 0,
                                stream, *args);
         } else {
             grid.x = (uint32_t)((args->n_chains * 2 + 255) / 256);
