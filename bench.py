@@ -279,7 +279,7 @@ def main():
     if rank == 0 and not args.no_cpu_baseline:
         lib_o = oracle._lib
         cores = os.cpu_count() or 1
-        sample_n = 128  # blocks; ~10-30 s of CPU work at 1 MiB blocks
+        sample_n = 1024  # blocks; tens of CPU-core-seconds at 1 MiB blocks
         log(f"[bench] cpu baseline: oracle, {sample_n} blocks, {cores} threads")
         if is_decode:
             el = lib_o.mo_cpu_reconstruct_bench(d, p, bs, sample_n,
