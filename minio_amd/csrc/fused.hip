@@ -97,23 +97,24 @@ __device__ __forceinline__ uint64_t shfl_x(uint64_t v, int mask) {
 
 } // namespace fused
 
-template <int D, int P, const uint8_t (&MAT)[P][D]>
+template <int D, int P, const uint8_t (&MAT)[P][D], int TILE>
 __global__ void __launch_bounds__(256) fused_encode_hh_kernel(FusedArgs a) {
     using namespace fused;
     constexpr int TOT = D + P;
     constexpr int G = 256 / (2 * TOT); /* blocks per workgroup */
-    constexpr int ROW = 272;           /* 256-B tile row + 16-B bank skew */
-    constexpr int NL = (G * D * 16 + 255) / 256; /* staged loads per lane */
+    constexpr int ROW = TILE + 16;     /* tile row + 16-B bank skew */
+    constexpr int TC = TILE / 16;      /* 16-B slots per tile row */
+    constexpr int NL = (G * D * TC + 255) / 256; /* staged loads per lane */
     __shared__ uint8_t lds[G * TOT * ROW];
 
     const int tid = threadIdx.x;
     const int64_t b0 = (int64_t)blockIdx.x * G;
     const int64_t S = a.shard_len;
     const int64_t stride = a.row_stride;
-    const int64_t n_iter = (S + 255) / 256;
+    const int64_t n_iter = (S + TILE - 1) / TILE;
     const int64_t full_pkts = S / 32;          /* whole 32-B packets */
     const int mod32 = (int)(S % 32);
-    const int64_t tail_tile = mod32 ? (S - mod32) / 256 : -1;
+    const int64_t tail_tile = mod32 ? (S - mod32) / TILE : -1;
 
     /* hash role */
     const int cp = tid >> 1;
@@ -145,11 +146,11 @@ __global__ void __launch_bounds__(256) fused_encode_hh_kernel(FusedArgs a) {
 #pragma unroll
         for (int l = 0; l < NL; l++) {
             const int e = tid + 256 * l;
-            if (NL * 256 > G * D * 16 && e >= G * D * 16) continue;
-            const int g = e / (D * 16);
-            const int r = e % (D * 16);
-            const int k = r / 16;
-            const int o = r % 16;
+            if (NL * 256 > G * D * TC && e >= G * D * TC) continue;
+            const int g = e / (D * TC);
+            const int r = e % (D * TC);
+            const int k = r / TC;
+            const int o = r % TC;
             const int64_t off = tbase + (int64_t)o * 16;
             if (b0 + g < a.n && off + 16 <= stride) {
                 st[l] = *(const uint4 *)(a.data +
@@ -162,30 +163,30 @@ __global__ void __launch_bounds__(256) fused_encode_hh_kernel(FusedArgs a) {
 
     load_tile(0);
     for (int64_t it = 0; it < n_iter; it++) {
-        const int64_t tbase = it * 256;
+        const int64_t tbase = it * TILE;
         __syncthreads(); /* previous tile fully consumed */
         /* A: staged registers -> LDS */
 #pragma unroll
         for (int l = 0; l < NL; l++) {
             const int e = tid + 256 * l;
-            if (NL * 256 > G * D * 16 && e >= G * D * 16) continue;
-            const int g = e / (D * 16);
-            const int r = e % (D * 16);
-            const int k = r / 16;
-            const int o = r % 16;
+            if (NL * 256 > G * D * TC && e >= G * D * TC) continue;
+            const int g = e / (D * TC);
+            const int r = e % (D * TC);
+            const int k = r / TC;
+            const int o = r % TC;
             *(uint4 *)&lds[(g * TOT + k) * ROW + o * 16] = st[l];
         }
-        if (it + 1 < n_iter) load_tile(tbase + 256); /* fly during B+C */
+        if (it + 1 < n_iter) load_tile(tbase + TILE); /* fly during B+C */
         __syncthreads(); /* data tile visible */
 
         /* B: GF parity for the tile.  One thread per 16-B column computes
          * ALL P parity rows (the doubling ladder is shared across rows and
          * every matrix index stays compile-time — a per-(row,column)
          * split would re-run the ladder P times and turn MAT[i][k] into a
-         * runtime load + branch). G*16 columns <= 256 threads. */
-        if (tid < G * 16) {
-            const int g = tid / 16;
-            const int o = tid % 16;
+         * runtime load + branch). */
+        for (int col = tid; col < G * TC; col += 256) {
+            const int g = col / TC;
+            const int o = col % TC;
             if (b0 + g < a.n) {
                 uint4 acc[P];
 #pragma unroll
@@ -228,7 +229,7 @@ __global__ void __launch_bounds__(256) fused_encode_hh_kernel(FusedArgs a) {
         if (chain_act) {
             const uint8_t *row = &lds[(cg * TOT + cs) * ROW + 16 * h];
             int pk = (int)(full_pkts - tbase / 32);
-            if (pk > 8) pk = 8;
+            if (pk > TILE / 32) pk = TILE / 32;
             for (int t = 0; t < pk; t++) {
                 uint4 q = *(const uint4 *)(row + 32 * t);
                 hh2_update(s, (uint64_t)q.x | ((uint64_t)q.y << 32),
@@ -307,14 +308,22 @@ extern "C" hipError_t mec_launch_fused_encode_hh(int d, int p,
      * kept for experimentation, off by default */
     static const char *env = getenv("MEC_FUSED");
     static const bool enabled = env && atoi(env) != 0;
+    static const char *envt = getenv("MEC_FUSED_TILE");
+    static const int tile = envt ? atoi(envt) : 512;
     if (!enabled) return hipErrorNotSupported;
     dim3 blk(256);
 #define X(D, P)                                                              \
     if (d == D && p == P) {                                                  \
         constexpr int G = 256 / (2 * (D + P));                               \
         dim3 grid((uint32_t)((args->n + G - 1) / G));                        \
-        hipLaunchKernelGGL((fused_encode_hh_kernel<D, P, MAT_##D##_##P>),    \
-                           grid, blk, 0, stream, *args);                     \
+        if (tile >= 512)                                                     \
+            hipLaunchKernelGGL(                                              \
+                (fused_encode_hh_kernel<D, P, MAT_##D##_##P, 512>), grid,    \
+                blk, 0, stream, *args);                                      \
+        else                                                                 \
+            hipLaunchKernelGGL(                                              \
+                (fused_encode_hh_kernel<D, P, MAT_##D##_##P, 256>), grid,    \
+                blk, 0, stream, *args);                                      \
         return hipGetLastError();                                            \
     }
     MEC_SPECIALIZED_GEOS(X)
