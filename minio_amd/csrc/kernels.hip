@@ -122,7 +122,7 @@ __global__ void __launch_bounds__(256) gf_encode_kernel(GfEncArgs a) {
  * (src = surviving rows, out = missing rows) via row index lists.
  * E = number of output rows (template so accumulators stay in registers).
  */
-template <int E>
+template <int E, bool MASKED>
 __global__ void __launch_bounds__(256) gf_matmul_kernel(GfMatmulArgs a) {
     const int b = blockIdx.y; /* batch item */
     const int64_t cols = (a.shard_len + 15) >> 4;
@@ -141,16 +141,32 @@ __global__ void __launch_bounds__(256) gf_matmul_kernel(GfMatmulArgs a) {
             uint4 pw = *(const uint4 *)(sbase +
                                         (int64_t)a.src_rows[k] * a.row_stride +
                                         j);
-            /* wave-uniform coefficient bits -> scalar branches */
             uint32_t cb[E];
 #pragma unroll
             for (int i = 0; i < E; i++) cb[i] = a.mat[i * MEC_KMAX_D + k];
+            uint32_t any = 0;
+#pragma unroll
+            for (int i = 0; i < E; i++) any |= cb[i];
 #pragma unroll
             for (int bit = 0; bit < 8; bit++) {
+                if (!(any >> bit)) break; /* uniform: no higher bits set */
                 if (bit) gf2x4(pw);
+                if (MASKED) {
+                    /* branch-free: broadcast scalar mask per (row,bit) */
 #pragma unroll
-                for (int i = 0; i < E; i++)
-                    if (cb[i] & (1u << bit)) xor4(acc[i], pw);
+                    for (int i = 0; i < E; i++) {
+                        uint32_t m = 0u - ((cb[i] >> bit) & 1u);
+                        acc[i].x ^= pw.x & m;
+                        acc[i].y ^= pw.y & m;
+                        acc[i].z ^= pw.z & m;
+                        acc[i].w ^= pw.w & m;
+                    }
+                } else {
+                    /* wave-uniform coefficient bits -> scalar branches */
+#pragma unroll
+                    for (int i = 0; i < E; i++)
+                        if (cb[i] & (1u << bit)) xor4(acc[i], pw);
+                }
             }
         }
 #pragma unroll
@@ -825,10 +841,16 @@ hipError_t mec_launch_gf_matmul(const GfMatmulArgs *args, int n_dst, int n,
     if (blocks_x < 1) blocks_x = 1;
     dim3 grid((uint32_t)blocks_x, n);
     dim3 blk(256);
+    static const char *envm = getenv("MEC_GFM_MASKED");
+    static const bool masked = envm && atoi(envm) != 0;
 #define CASE(E)                                                              \
     case E:                                                                  \
-        hipLaunchKernelGGL(gf_matmul_kernel<E>, grid, blk, 0, stream,        \
-                           *args);                                           \
+        if (masked)                                                          \
+            hipLaunchKernelGGL((gf_matmul_kernel<E, true>), grid, blk, 0,    \
+                               stream, *args);                               \
+        else                                                                 \
+            hipLaunchKernelGGL((gf_matmul_kernel<E, false>), grid, blk, 0,   \
+                               stream, *args);                               \
         break;
     switch (n_dst) {
         CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
