@@ -59,11 +59,6 @@ Geo make_geo(int d, int p, int64_t block_size, int64_t total_length) {
     return g;
 }
 
-/* shard size of block `b` */
-int64_t blk_S(const Geo &g, int64_t b) {
-    return (b == g.n_blocks - 1) ? g.last_S : g.S;
-}
-
 /* offset of block b's shard inside the per-drive stream */
 int64_t stream_off(const Geo &g, int64_t b, int algo) {
     int hsz = (algo == MEC_BITROT_HIGHWAYHASH256S) ? 32 : 0;
