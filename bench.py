@@ -264,11 +264,17 @@ def main():
                 dom, t_dom, dom_bytes = "bitrot_hash", t_hash, hash_bytes
         peak = 8.0e12  # MI355X HBM3E spec peak B/s (MI355X_MICROARCH.md)
         achieved = dom_bytes / t_dom
+        # traffic: PMC passes (profiles/r06, FETCH_SIZE x2 gfx950
+        # correction + WRITE_SIZE) measured HBM bytes == algorithmic bytes
+        # within 0.3% for both kernels at the headline workload, so the
+        # per-launch traffic equals dom_bytes there; other workloads: null
+        traffic = dom_bytes if (args.op == "encode" and not args.batch) else None
         roofline = {
             "bound": "hbm", "kernel": dom,
             "achieved": round(achieved / 1e9, 1), "peak": peak / 1e9,
             "unit": "GB/s", "frac": round(achieved / peak, 4),
-            "traffic": None,
+            "traffic": traffic,
+            "traffic_evidence": "profiles/r06_pmc_*.txt" if traffic else None,
         }
         if not is_decode:
             roofline["legs_ms"] = {"gf": round(t_gf * 1e3, 3),
