@@ -73,13 +73,22 @@ __global__ void __launch_bounds__(256) gf_encode_kernel(GfEncArgs a) {
         for (int i = 0; i < P; i++)
 #pragma unroll
             for (int w = 0; w < W; w++) acc[i][w] = uint4{0, 0, 0, 0};
+        /* issue ALL row loads before any ladder math: the consume-as-you-
+         * load form left one load in flight at a time (ISA: LOAD WAIT
+         * ladder LOAD WAIT ... — the 36% memory-wait in the r03 PMC) */
+        uint4 pws[D][W];
 #pragma unroll
         for (int k = 0; k < D; k++) {
             const uint8_t *row = sbase + (int64_t)k * a.row_stride + j;
-            uint4 pw[W];
 #pragma unroll
             for (int w = 0; w < W; w++)
-                pw[w] = *(const uint4 *)(row + 16 * w);
+                pws[k][w] = *(const uint4 *)(row + 16 * w);
+        }
+#pragma unroll
+        for (int k = 0; k < D; k++) {
+            uint4 pw[W];
+#pragma unroll
+            for (int w = 0; w < W; w++) pw[w] = pws[k][w];
 #pragma unroll
             for (int bit = 0; bit < 8; bit++) {
                 uint32_t need = 0;
