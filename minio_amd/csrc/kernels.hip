@@ -190,111 +190,6 @@ __global__ void __launch_bounds__(256) gf_matmul_kernel(GfMatmulArgs a) {
  * magic key passed from host; pinned by tests/golden/bitrot_selftest.json).
  */
 
-struct HHState {
-    uint64_t v0[4], v1[4], mul0[4], mul1[4];
-};
-
-__device__ __forceinline__ void hh_zma(uint64_t v1, uint64_t v0,
-                                       uint64_t &add1, uint64_t &add0) {
-    add0 += (((v0 & 0xff000000ull) | (v1 & 0xff00000000ull)) >> 24) |
-            (((v0 & 0xff0000000000ull) | (v1 & 0xff000000000000ull)) >> 16) |
-            (v0 & 0xff0000ull) | ((v0 & 0xff00ull) << 32) |
-            ((v1 & 0xff00000000000000ull) >> 8) | (v0 << 56);
-    add1 += (((v1 & 0xff000000ull) | (v0 & 0xff00000000ull)) >> 24) |
-            (v1 & 0xff0000ull) | ((v1 & 0xff0000000000ull) >> 16) |
-            ((v1 & 0xff00ull) << 24) | ((v0 & 0xff000000000000ull) >> 8) |
-            ((v1 & 0xffull) << 48) | (v0 & 0xff00000000000000ull);
-}
-
-__device__ __forceinline__ void hh_update(HHState &s, const uint64_t lanes[4]) {
-#pragma unroll
-    for (int i = 0; i < 4; i++) {
-        s.v1[i] += s.mul0[i] + lanes[i];
-        s.mul0[i] ^= (s.v1[i] & 0xffffffffull) * (s.v0[i] >> 32);
-        s.v0[i] += s.mul1[i];
-        s.mul1[i] ^= (s.v0[i] & 0xffffffffull) * (s.v1[i] >> 32);
-    }
-    hh_zma(s.v1[1], s.v1[0], s.v0[1], s.v0[0]);
-    hh_zma(s.v1[3], s.v1[2], s.v0[3], s.v0[2]);
-    hh_zma(s.v0[1], s.v0[0], s.v1[1], s.v1[0]);
-    hh_zma(s.v0[3], s.v0[2], s.v1[3], s.v1[2]);
-}
-
-__device__ __forceinline__ void hh_reset(HHState &s, const uint64_t key[4]) {
-    const uint64_t init0[4] = {0xdbe6d5d5fe4cce2full, 0xa4093822299f31d0ull,
-                               0x13198a2e03707344ull, 0x243f6a8885a308d3ull};
-    const uint64_t init1[4] = {0x3bd39e10cb0ef593ull, 0xc0acf169b5f18a8cull,
-                               0xbe5466cf34e90c6cull, 0x452821e638d01377ull};
-#pragma unroll
-    for (int i = 0; i < 4; i++) {
-        s.mul0[i] = init0[i];
-        s.mul1[i] = init1[i];
-        s.v0[i] = init0[i] ^ key[i];
-        s.v1[i] = init1[i] ^ ((key[i] >> 32) | (key[i] << 32));
-    }
-}
-
-__device__ __forceinline__ void hh_finalize256(HHState &s, uint64_t out[4]) {
-#pragma unroll 1
-    for (int r = 0; r < 10; r++) {
-        uint64_t perm[4];
-        perm[0] = (s.v0[2] >> 32) | (s.v0[2] << 32);
-        perm[1] = (s.v0[3] >> 32) | (s.v0[3] << 32);
-        perm[2] = (s.v0[0] >> 32) | (s.v0[0] << 32);
-        perm[3] = (s.v0[1] >> 32) | (s.v0[1] << 32);
-        hh_update(s, perm);
-    }
-    uint64_t a2, a3;
-    a3 = (s.v1[1] + s.mul1[1]) & 0x3fffffffffffffffull;
-    a2 = s.v1[0] + s.mul1[0];
-    out[1] = (s.v0[1] + s.mul0[1]) ^ ((a3 << 1) | (a2 >> 63)) ^
-             ((a3 << 2) | (a2 >> 62));
-    out[0] = (s.v0[0] + s.mul0[0]) ^ (a2 << 1) ^ (a2 << 2);
-    a3 = (s.v1[3] + s.mul1[3]) & 0x3fffffffffffffffull;
-    a2 = s.v1[2] + s.mul1[2];
-    out[3] = (s.v0[3] + s.mul0[3]) ^ ((a3 << 1) | (a2 >> 63)) ^
-             ((a3 << 2) | (a2 >> 62));
-    out[2] = (s.v0[2] + s.mul0[2]) ^ (a2 << 1) ^ (a2 << 2);
-}
-
-/* Map chain index -> (shard pointer, sum slot) for the fused encode
- * layout.  Sums always use the n x (d+p) layout regardless of mode, so
- * data-only + parity-only launches compose to the single-launch result. */
-__device__ __forceinline__ const uint8_t *chain_ptr(const HashArgs &a,
-                                                    int64_t chain,
-                                                    int64_t &sum_idx) {
-    if (a.parity == nullptr) {
-        sum_idx = chain;
-        return a.data + chain * a.row_stride;
-    }
-    const int total = a.d + a.p;
-    if (a.mode == MEC_HASH_DATA) {
-        const int64_t b = chain / a.d;
-        const int s = (int)(chain % a.d);
-        sum_idx = b * total + s;
-        return a.data + (b * a.d + s) * a.row_stride;
-    }
-    if (a.mode == MEC_HASH_PARITY) {
-        const int64_t b = chain / a.p;
-        const int s = (int)(chain % a.p);
-        sum_idx = b * total + a.d + s;
-        return a.parity + (b * a.p + s) * a.row_stride;
-    }
-    const int64_t b = chain / total;
-    const int s = (int)(chain % total);
-    sum_idx = chain;
-    if (s < a.d)
-        return a.data + (b * a.d + s) * a.row_stride;
-    return a.parity + (b * a.p + (s - a.d)) * a.row_stride;
-}
-
-/* 4-lane HighwayHash: one chain per 4 consecutive GPU lanes, lane li owns
- * HighwayHash lane li of the state.  Zipper merges and the finalization
- * permute exchange partner words via __shfl_xor inside the quad.  4x the
- * chain parallelism of a per-lane design and 1/4 the serial work per lane —
- * hashing is sequential per shard, so chains x lanes is the only
- * parallelism available (SURVEY.md §7 hard part (b)). */
-
 /* The zipper merge is a pure byte permutation of the pair (the reference
  * implements it with pshufb); on CDNA4 that is v_perm_b32: 3 perms + 1 or
  * per output word instead of a ~17-op shift/mask tree.
