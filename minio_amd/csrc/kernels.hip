@@ -190,6 +190,37 @@ __global__ void __launch_bounds__(256) gf_matmul_kernel(GfMatmulArgs a) {
  * magic key passed from host; pinned by tests/golden/bitrot_selftest.json).
  */
 
+/* Map chain index -> (shard pointer, sum slot) for the fused encode
+ * layout.  Sums always use the n x (d+p) layout regardless of mode, so
+ * data-only + parity-only launches compose to the single-launch result. */
+__device__ __forceinline__ const uint8_t *chain_ptr(const HashArgs &a,
+                                                    int64_t chain,
+                                                    int64_t &sum_idx) {
+    if (a.parity == nullptr) {
+        sum_idx = chain;
+        return a.data + chain * a.row_stride;
+    }
+    const int total = a.d + a.p;
+    if (a.mode == MEC_HASH_DATA) {
+        const int64_t b = chain / a.d;
+        const int s = (int)(chain % a.d);
+        sum_idx = b * total + s;
+        return a.data + (b * a.d + s) * a.row_stride;
+    }
+    if (a.mode == MEC_HASH_PARITY) {
+        const int64_t b = chain / a.p;
+        const int s = (int)(chain % a.p);
+        sum_idx = b * total + a.d + s;
+        return a.parity + (b * a.p + s) * a.row_stride;
+    }
+    const int64_t b = chain / total;
+    const int s = (int)(chain % total);
+    sum_idx = chain;
+    if (s < a.d)
+        return a.data + (b * a.d + s) * a.row_stride;
+    return a.parity + (b * a.p + (s - a.d)) * a.row_stride;
+}
+
 /* The zipper merge is a pure byte permutation of the pair (the reference
  * implements it with pshufb); on CDNA4 that is v_perm_b32: 3 perms + 1 or
  * per output word instead of a ~17-op shift/mask tree.

@@ -321,3 +321,24 @@ def test_multiblock_object_stream():
         dmg = [None] * p + list(streams[p:])
         healed = e.heal_stream(dmg, total)
         assert healed == streams
+
+
+def test_encode_parity_large_generic_geometries():
+    # beyond the specialized list -> generic gf_matmul path (d<=32,
+    # total<=40 per MEC_KMAX_*; the reference allows up to 256 shards,
+    # deployments cap at 16 drives/set, docs/distributed/DESIGN.md:44-52)
+    for d, p in [(20, 6), (32, 8), (17, 5)]:
+        bs = 64 * 1024
+        data = rnd(bs, SEED + d * 7 + p)
+        with minio_amd.Erasure(d, p, bs) as e:
+            shards, sums = e.encode_batch(data, bs, 2,
+                                          minio_amd.HIGHWAYHASH256S)
+        ors = oracle.RS(d, p)
+        want = ors.encode_data(data)
+        assert shards[0] == want, f"d={d} p={p}"
+        for s, sh in enumerate(want):
+            assert sums[0][s] == oracle.bitrot_sum(oracle.HIGHWAYHASH256S, sh)
+        # and reconstruct with p erasures through the generic path
+        damaged = [None] * p + want[p:]
+        rec = e.decode_data_and_parity_blocks(damaged)
+        assert rec == want
