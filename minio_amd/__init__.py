@@ -187,6 +187,11 @@ class Erasure:
     def __exit__(self, *exc):
         self.close()
 
+    def _ck(self):
+        if not self._ctx:
+            raise MecError(6, "context is closed")
+        return self._ctx
+
     # -- size math (instance mirrors) --
     def shard_size(self) -> int:
         return shard_size(self.block_size, self.d)
@@ -219,7 +224,7 @@ class Erasure:
         if algo is not None:
             sums_buf = ctypes.create_string_buffer(
                 n * (self.d + self.p) * _HASH_SIZE[algo])
-        _check(_lib.mec_encode_batch(self._ctx, n, blocks, block_len, parity,
+        _check(_lib.mec_encode_batch(self._ck(), n, blocks, block_len, parity,
                                      algo or 0, sums_buf))
         out = []
         hs = _HASH_SIZE[algo] if algo is not None else 0
@@ -264,7 +269,7 @@ class Erasure:
         for i, s in enumerate(shards):
             if s:
                 buf[i * per:(i + 1) * per] = s
-        _check(_lib.mec_reconstruct_batch(self._ctx, 1, buf, present, per,
+        _check(_lib.mec_reconstruct_batch(self._ck(), 1, buf, present, per,
                                           1 if data_only else 0))
         out = []
         for i in range(total):
@@ -288,7 +293,7 @@ class Erasure:
         ws = None
         if algo != HIGHWAYHASH256S:
             ws = ctypes.create_string_buffer(total * _HASH_SIZE[algo])
-        _check(_lib.mec_encode_stream(self._ctx, src, len(src), algo, arr, ws))
+        _check(_lib.mec_encode_stream(self._ck(), src, len(src), algo, arr, ws))
         streams = [b.raw[:fsz] for b in bufs]
         sums = None
         if ws is not None:
@@ -308,7 +313,7 @@ class Erasure:
             ws = b"".join(s if s else b"\0" * _HASH_SIZE[algo]
                           for s in whole_sums)
         dst = ctypes.create_string_buffer(max(length, 1))
-        _check(_lib.mec_decode_stream(self._ctx, arr, ws, algo, total_length,
+        _check(_lib.mec_decode_stream(self._ck(), arr, ws, algo, total_length,
                                       offset, length, dst))
         return dst.raw[:length]
 
@@ -331,7 +336,7 @@ class Erasure:
             else:
                 outs.append(None)
                 outp[i] = None
-        _check(_lib.mec_heal_stream(self._ctx, arr, algo, total_length, outp))
+        _check(_lib.mec_heal_stream(self._ck(), arr, algo, total_length, outp))
         return [
             drive_streams[i] if drive_streams[i] is not None
             else outs[i].raw[:fsz]
@@ -343,7 +348,7 @@ class Erasure:
                          msg_stride: int, n: int):
         hs = _HASH_SIZE[algo]
         out = ctypes.create_string_buffer(max(n * hs, 1))
-        _check(_lib.mec_bitrot_sum_batch(self._ctx, algo, n, msgs, msg_len,
+        _check(_lib.mec_bitrot_sum_batch(self._ck(), algo, n, msgs, msg_len,
                                          msg_stride, out))
         return [out.raw[i * hs:(i + 1) * hs] for i in range(n)]
 
@@ -354,7 +359,7 @@ class Erasure:
         ss = shard_size_ if shard_size_ is not None else self.shard_size()
         try:
             _check(_lib.mec_bitrot_verify_stream(
-                self._ctx, stream, len(stream), part_size, algo, want_sum, ss))
+                self._ck(), stream, len(stream), part_size, algo, want_sum, ss))
             return True
         except FileCorruptError:
             return False

@@ -331,6 +331,7 @@ mec_status mec_encode_batch_dev_pipe(mec_ctx *ctx, int n,
                                      const void *data_dev, int64_t block_len,
                                      void *parity_dev, int algo,
                                      void *sums_dev) {
+    if (!ctx) return MEC_ERR_INVALID_ARG;
     std::lock_guard<std::mutex> lk(ctx->mu);
     if (n <= 0 || block_len <= 0 || block_len > ctx->block_size ||
         sums_dev == nullptr || !hash_size(algo))
@@ -378,6 +379,7 @@ mec_status mec_encode_batch_dev_pipe(mec_ctx *ctx, int n,
 }
 
 mec_status mec_pipe_sync(mec_ctx *ctx) {
+    if (!ctx) return MEC_ERR_INVALID_ARG;
     std::lock_guard<std::mutex> lk(ctx->mu);
     HIP_TRY(hipSetDevice(ctx->device));
     HIP_TRY(hipStreamSynchronize(ctx->stream));
@@ -389,6 +391,7 @@ mec_status mec_encode_batch_dev_async(mec_ctx *ctx, int n,
                                       const void *data_dev, int64_t block_len,
                                       void *parity_dev, int algo,
                                       void *sums_dev) {
+    if (!ctx) return MEC_ERR_INVALID_ARG;
     std::lock_guard<std::mutex> lk(ctx->mu);
     return encode_dev_locked(ctx, n, data_dev, block_len, parity_dev, algo,
                              sums_dev);
@@ -397,6 +400,7 @@ mec_status mec_encode_batch_dev_async(mec_ctx *ctx, int n,
 mec_status mec_encode_batch_dev(mec_ctx *ctx, int n, const void *data_dev,
                                 int64_t block_len, void *parity_dev, int algo,
                                 void *sums_dev) {
+    if (!ctx) return MEC_ERR_INVALID_ARG;
     std::lock_guard<std::mutex> lk(ctx->mu);
     mec_status s = encode_dev_locked(ctx, n, data_dev, block_len, parity_dev,
                                      algo, sums_dev);
@@ -408,6 +412,7 @@ mec_status mec_encode_batch_dev(mec_ctx *ctx, int n, const void *data_dev,
 mec_status mec_encode_batch(mec_ctx *ctx, int n, const uint8_t *data,
                             int64_t block_len, uint8_t *parity, int algo,
                             uint8_t *sums) {
+    if (!ctx) return MEC_ERR_INVALID_ARG;
     if (n <= 0 || block_len <= 0 || block_len > ctx->block_size)
         return MEC_ERR_INVALID_ARG;
     std::lock_guard<std::mutex> lk(ctx->mu);
@@ -514,6 +519,7 @@ mec_status mec_reconstruct_batch_dev_async(mec_ctx *ctx, int n,
                                            void *shards_dev,
                                            const uint8_t *present,
                                            int64_t shard_len, int data_only) {
+    if (!ctx) return MEC_ERR_INVALID_ARG;
     std::lock_guard<std::mutex> lk(ctx->mu);
     return reconstruct_dev_locked(ctx, n, shards_dev, present, shard_len,
                                   data_only);
@@ -522,6 +528,7 @@ mec_status mec_reconstruct_batch_dev_async(mec_ctx *ctx, int n,
 mec_status mec_reconstruct_batch_dev(mec_ctx *ctx, int n, void *shards_dev,
                                      const uint8_t *present,
                                      int64_t shard_len, int data_only) {
+    if (!ctx) return MEC_ERR_INVALID_ARG;
     std::lock_guard<std::mutex> lk(ctx->mu);
     mec_status s = reconstruct_dev_locked(ctx, n, shards_dev, present,
                                           shard_len, data_only);
@@ -533,6 +540,7 @@ mec_status mec_reconstruct_batch_dev(mec_ctx *ctx, int n, void *shards_dev,
 mec_status mec_reconstruct_batch(mec_ctx *ctx, int n, uint8_t *shards,
                                  const uint8_t *present, int64_t shard_len,
                                  int data_only) {
+    if (!ctx) return MEC_ERR_INVALID_ARG;
     if (n <= 0 || shard_len <= 0 || shard_len > ctx->stride)
         return MEC_ERR_INVALID_ARG;
     std::lock_guard<std::mutex> lk(ctx->mu);
@@ -575,6 +583,7 @@ mec_status mec_reconstruct_batch(mec_ctx *ctx, int n, uint8_t *shards,
 mec_status mec_bitrot_sum_batch_dev(mec_ctx *ctx, int algo, int n,
                                     const void *msgs_dev, int64_t msg_len,
                                     int64_t msg_stride, void *sums_dev) {
+    if (!ctx) return MEC_ERR_INVALID_ARG;
     if (n <= 0 || msg_len < 0 || !hash_size(algo)) return MEC_ERR_INVALID_ARG;
     std::lock_guard<std::mutex> lk(ctx->mu);
     HIP_TRY(hipSetDevice(ctx->device));
@@ -594,6 +603,7 @@ mec_status mec_bitrot_sum_batch_dev(mec_ctx *ctx, int algo, int n,
 mec_status mec_bitrot_sum_batch(mec_ctx *ctx, int algo, int n,
                                 const uint8_t *msgs, int64_t msg_len,
                                 int64_t msg_stride, uint8_t *sums) {
+    if (!ctx) return MEC_ERR_INVALID_ARG;
     if (n <= 0 || msg_len < 0 || !hash_size(algo)) return MEC_ERR_INVALID_ARG;
     const int hsz = hash_size(algo);
     std::lock_guard<std::mutex> lk(ctx->mu);
@@ -651,6 +661,7 @@ mec_status mec_bitrot_verify_batch(mec_ctx *ctx, int algo, int n,
 /* ---- device memory + timing helpers ------------------------------------ */
 
 mec_status mec_dev_alloc(mec_ctx *ctx, size_t bytes, void **out) {
+    if (!ctx) return MEC_ERR_INVALID_ARG;
     HIP_TRY(hipSetDevice(ctx->device));
     HIP_TRY(hipMalloc(out, bytes));
     return MEC_OK;
@@ -663,6 +674,7 @@ void mec_dev_free(mec_ctx *ctx, void *ptr) {
 
 mec_status mec_memcpy_h2d(mec_ctx *ctx, void *dst_dev, const void *src,
                           size_t bytes) {
+    if (!ctx) return MEC_ERR_INVALID_ARG;
     HIP_TRY(hipSetDevice(ctx->device));
     HIP_TRY(hipMemcpyAsync(dst_dev, src, bytes, hipMemcpyHostToDevice,
                            ctx->stream));
@@ -672,6 +684,7 @@ mec_status mec_memcpy_h2d(mec_ctx *ctx, void *dst_dev, const void *src,
 
 mec_status mec_memcpy_d2h(mec_ctx *ctx, void *dst, const void *src_dev,
                           size_t bytes) {
+    if (!ctx) return MEC_ERR_INVALID_ARG;
     HIP_TRY(hipSetDevice(ctx->device));
     HIP_TRY(hipMemcpyAsync(dst, src_dev, bytes, hipMemcpyDeviceToHost,
                            ctx->stream));
@@ -681,6 +694,7 @@ mec_status mec_memcpy_d2h(mec_ctx *ctx, void *dst, const void *src_dev,
 
 mec_status mec_memset_dev(mec_ctx *ctx, void *dst_dev, int value,
                           size_t bytes) {
+    if (!ctx) return MEC_ERR_INVALID_ARG;
     HIP_TRY(hipSetDevice(ctx->device));
     HIP_TRY(hipMemsetAsync(dst_dev, value, bytes, ctx->stream));
     HIP_TRY(hipStreamSynchronize(ctx->stream));
@@ -688,18 +702,21 @@ mec_status mec_memset_dev(mec_ctx *ctx, void *dst_dev, int value,
 }
 
 mec_status mec_stream_sync(mec_ctx *ctx) {
+    if (!ctx) return MEC_ERR_INVALID_ARG;
     HIP_TRY(hipSetDevice(ctx->device));
     HIP_TRY(hipStreamSynchronize(ctx->stream));
     return MEC_OK;
 }
 
 mec_status mec_timer_start(mec_ctx *ctx) {
+    if (!ctx) return MEC_ERR_INVALID_ARG;
     HIP_TRY(hipSetDevice(ctx->device));
     HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
     return MEC_OK;
 }
 
 mec_status mec_timer_stop(mec_ctx *ctx, float *ms_out) {
+    if (!ctx) return MEC_ERR_INVALID_ARG;
     HIP_TRY(hipSetDevice(ctx->device));
     HIP_TRY(hipEventRecord(ctx->ev_stop, ctx->stream));
     HIP_TRY(hipEventSynchronize(ctx->ev_stop));

@@ -333,12 +333,17 @@ def test_encode_parity_large_generic_geometries():
         with minio_amd.Erasure(d, p, bs) as e:
             shards, sums = e.encode_batch(data, bs, 2,
                                           minio_amd.HIGHWAYHASH256S)
-        ors = oracle.RS(d, p)
-        want = ors.encode_data(data[:bs])
-        assert shards[0] == want, f"d={d} p={p}"
-        for s, sh in enumerate(want):
-            assert sums[0][s] == oracle.bitrot_sum(oracle.HIGHWAYHASH256S, sh)
-        # and reconstruct with p erasures through the generic path
-        damaged = [None] * p + want[p:]
-        rec = e.decode_data_and_parity_blocks(damaged)
-        assert rec == want
+            ors = oracle.RS(d, p)
+            want = ors.encode_data(data[:bs])
+            assert shards[0] == want, f"d={d} p={p}"
+            for s, sh in enumerate(want):
+                assert sums[0][s] == oracle.bitrot_sum(
+                    oracle.HIGHWAYHASH256S, sh)
+            # and reconstruct with p erasures through the generic path
+            damaged = [None] * p + want[p:]
+            rec = e.decode_data_and_parity_blocks(damaged)
+            assert rec == want
+        # use-after-close must raise, not crash (the segfault this test
+        # originally tripped: a NULL ctx reached the C-ABI)
+        with pytest.raises(minio_amd.MecError):
+            e.decode_data_and_parity_blocks(damaged)
