@@ -65,38 +65,30 @@ __global__ void __launch_bounds__(256) gf_encode_kernel(GfEncArgs a) {
     const uint8_t *__restrict__ sbase = a.data + (int64_t)b * D * a.row_stride;
     uint8_t *__restrict__ obase = a.parity + (int64_t)b * P * a.row_stride;
 
-    /* software pipeline across grid-stride columns: iteration c's ladder
-     * math runs while c+stride's row loads are in flight.  (A plain
-     * hoist of the loads gets re-fused by the scheduler: the ISA showed
-     * LOAD-WAIT-ladder x8 with one load in flight — r03's 36% mem-wait.) */
-    const int64_t gstride = (int64_t)gridDim.x * blockDim.x;
-    int64_t c = blockIdx.x * blockDim.x + threadIdx.x;
-    if (c >= cols) return;
-    uint4 cur[D][W], nxt[D][W];
-#define GF_LOAD(DST, COL)                                                    \
-    {                                                                        \
-        const int64_t jj = (COL) * (16 * W);                                 \
-        _Pragma("unroll") for (int k = 0; k < D; k++) {                      \
-            const uint8_t *row = sbase + (int64_t)k * a.row_stride + jj;     \
-            _Pragma("unroll") for (int w = 0; w < W; w++)                    \
-                DST[k][w] = *(const uint4 *)(row + 16 * w);                  \
-        }                                                                    \
-    }
-    GF_LOAD(cur, c)
-    for (; c < cols; c += gstride) {
+    for (int64_t c = blockIdx.x * blockDim.x + threadIdx.x; c < cols;
+         c += (int64_t)gridDim.x * blockDim.x) {
         const int64_t j = c * (16 * W);
-        const bool have_next = c + gstride < cols;
-        if (have_next) GF_LOAD(nxt, c + gstride)
         uint4 acc[P][W];
 #pragma unroll
         for (int i = 0; i < P; i++)
 #pragma unroll
             for (int w = 0; w < W; w++) acc[i][w] = uint4{0, 0, 0, 0};
+        /* issue ALL row loads before any ladder math: the consume-as-you-
+         * load form left one load in flight at a time (ISA: LOAD WAIT
+         * ladder LOAD WAIT ... — the 36% memory-wait in the r03 PMC) */
+        uint4 pws[D][W];
+#pragma unroll
+        for (int k = 0; k < D; k++) {
+            const uint8_t *row = sbase + (int64_t)k * a.row_stride + j;
+#pragma unroll
+            for (int w = 0; w < W; w++)
+                pws[k][w] = *(const uint4 *)(row + 16 * w);
+        }
 #pragma unroll
         for (int k = 0; k < D; k++) {
             uint4 pw[W];
 #pragma unroll
-            for (int w = 0; w < W; w++) pw[w] = cur[k][w];
+            for (int w = 0; w < W; w++) pw[w] = pws[k][w];
 #pragma unroll
             for (int bit = 0; bit < 8; bit++) {
                 uint32_t need = 0;
@@ -129,14 +121,7 @@ __global__ void __launch_bounds__(256) gf_encode_kernel(GfEncArgs a) {
                 }
             }
         }
-        if (have_next) {
-#pragma unroll
-            for (int k = 0; k < D; k++)
-#pragma unroll
-                for (int w = 0; w < W; w++) cur[k][w] = nxt[k][w];
-        }
     }
-#undef GF_LOAD
 }
 
 /* ---- generic GF matrix-multiply over shard rows ------------------------
