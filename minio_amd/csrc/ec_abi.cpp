@@ -256,7 +256,13 @@ static mec_status encode_dev_locked(mec_ctx *ctx, int n, const void *data_dev,
         fa.shard_len = S_call;
         fa.n = n;
         memcpy(fa.key, kMagicHHKey, 32);
-        hipError_t he = mec_launch_fused_encode_hh(d, p, &fa, ctx->stream);
+        hipError_t he = mec_launch_fused2_encode_hh(d, p, &fa, ctx->stream);
+        if (he == hipSuccess) return MEC_OK;
+        if (he != hipErrorNotSupported) {
+            set_err("fused2_encode_hh", he);
+            return MEC_ERR_HIP;
+        }
+        he = mec_launch_fused_encode_hh(d, p, &fa, ctx->stream);
         if (he == hipSuccess) return MEC_OK;
         if (he != hipErrorNotSupported) {
             set_err("fused_encode_hh", he);

@@ -66,6 +66,8 @@ struct FusedArgs {
 extern "C" {
 hipError_t mec_launch_fused_encode_hh(int d, int p, const FusedArgs *args,
                                       hipStream_t stream);
+hipError_t mec_launch_fused2_encode_hh(int d, int p, const FusedArgs *args,
+                                       hipStream_t stream);
 hipError_t mec_launch_gf_matmul(const GfMatmulArgs *args, int n_dst, int n,
                                 hipStream_t stream);
 hipError_t mec_launch_gf_encode_spec(int d, int p, const GfEncArgs *args,
