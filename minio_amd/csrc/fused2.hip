@@ -232,6 +232,9 @@ __global__ void __launch_bounds__(512) fused2_encode_hh_kernel(FusedArgs a) {
         __hip_atomic_load(&flags[slot], __ATOMIC_ACQUIRE,
                           __HIP_MEMORY_SCOPE_WORKGROUP);
         if (act) {
+            /* chains are the latency wall: outprioritize the producer
+             * waves sharing this SIMD (T5) */
+            __builtin_amdgcn_s_setprio(1);
             const uint8_t *row =
                 &lds[slot * SLOT + (cg * TOT + cs) * ROW + 16 * h];
 #pragma unroll 4
@@ -240,6 +243,7 @@ __global__ void __launch_bounds__(512) fused2_encode_hh_kernel(FusedArgs a) {
                 hh2_update(s, (uint64_t)q.x | ((uint64_t)q.y << 32),
                            (uint64_t)q.z | ((uint64_t)q.w << 32));
             }
+            __builtin_amdgcn_s_setprio(0);
         }
         __builtin_amdgcn_s_waitcnt(0);
         if ((tid & 63) == 0) {
