@@ -807,6 +807,7 @@ hipError_t mec_launch_hash(int algo, const HashArgs *args,
         grid.x = (uint32_t)(((args->n_chains + 1) / 2 + 255) / 256);
         hipLaunchKernelGGL(sha256_batch_kernel<2>, grid, blk, 0, stream,
                            *args);
+
         break;
     case 2: /* HighwayHash256 */
     case 3: /* HighwayHash256S: 2 lanes/chain (zipper pairs).  32-aligned
