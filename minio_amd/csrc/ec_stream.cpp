@@ -340,6 +340,40 @@ mec_status mec_heal_stream(mec_ctx *ctx_, const uint8_t *const *drive_bufs,
     }
     if (n_present < d) return MEC_ERR_TOO_FEW_SHARDS;
 
+    /* verify-on-read (Heal reads through bitrot readers,
+     * cmd/erasure-decode.go:322 + bitrot-streaming.go:185-197): a corrupt
+     * shard is treated as missing for its block.  Whole-drive verification
+     * batched per drive; per-block masks grouped below. */
+    std::vector<uint8_t> shard_ok((size_t)g.n_blocks * g.total, 0);
+    for (int s = 0; s < g.total; s++) {
+        if (!mask[s]) continue;
+        bool ragged_last = g.last_S != g.S;
+        int64_t full_nb = ragged_last ? g.n_blocks - 1 : g.n_blocks;
+        if (full_nb > 0) {
+            std::vector<uint8_t> ok((size_t)full_nb);
+            std::vector<uint8_t> want((size_t)full_nb * hsz);
+            for (int64_t b = 0; b < full_nb; b++)
+                memcpy(want.data() + (size_t)b * hsz,
+                       drive_bufs[s] + b * (hsz + g.S), (size_t)hsz);
+            mec_status st = mec_bitrot_verify_batch(
+                ctx_, algo, (int)full_nb, drive_bufs[s] + hsz, g.S,
+                hsz + g.S, want.data(), ok.data());
+            if (st != MEC_OK) return st;
+            for (int64_t b = 0; b < full_nb; b++)
+                shard_ok[b * g.total + s] = ok[b];
+        }
+        if (ragged_last) {
+            const uint8_t *hp =
+                drive_bufs[s] + stream_off(g, g.n_blocks - 1, algo);
+            uint8_t ok = 0;
+            mec_status st = mec_bitrot_verify_batch(ctx_, algo, 1, hp + hsz,
+                                                    g.last_S,
+                                                    g.last_S + hsz, hp, &ok);
+            if (st != MEC_OK) return st;
+            shard_ok[(g.n_blocks - 1) * g.total + s] = ok;
+        }
+    }
+
     for (int pass = 0; pass < 2; pass++) {
         int64_t b0 = pass == 0 ? 0 : g.n_blocks - 1;
         int64_t n;
@@ -358,18 +392,41 @@ mec_status mec_heal_stream(mec_ctx *ctx_, const uint8_t *const *drive_bufs,
         std::vector<uint8_t> rows((size_t)n * g.total * S, 0);
         for (int64_t i = 0; i < n; i++)
             for (int s = 0; s < g.total; s++) {
-                if (!mask[s]) continue;
+                if (!shard_ok[(b0 + i) * g.total + s]) continue;
                 memcpy(rows.data() + ((size_t)i * g.total + s) * S,
                        drive_bufs[s] + stream_off(g, b0 + i, algo) + hsz,
                        (size_t)S);
             }
-        mec_status st = mec_reconstruct_batch(ctx_, (int)n, rows.data(),
-                                              mask.data(), S, 0);
-        if (st != MEC_OK) return st;
+        /* group blocks by their verified-present mask */
+        std::map<std::vector<uint8_t>, std::vector<int64_t>> groups;
+        for (int64_t i = 0; i < n; i++) {
+            std::vector<uint8_t> m(shard_ok.begin() + (b0 + i) * g.total,
+                                   shard_ok.begin() + (b0 + i + 1) * g.total);
+            groups[m].push_back(i);
+        }
+        for (auto &kv : groups) {
+            int np = 0;
+            for (int s = 0; s < g.total; s++) np += kv.first[s] != 0;
+            if (np < d) return MEC_ERR_FILE_CORRUPT;
+            if (np == g.total) continue;
+            int64_t gn = (int64_t)kv.second.size();
+            std::vector<uint8_t> grp((size_t)gn * g.total * S, 0);
+            for (int64_t t = 0; t < gn; t++)
+                memcpy(grp.data() + (size_t)t * g.total * S,
+                       rows.data() + (size_t)kv.second[t] * g.total * S,
+                       (size_t)g.total * S);
+            mec_status st = mec_reconstruct_batch(
+                ctx_, (int)gn, grp.data(), kv.first.data(), S, 0);
+            if (st != MEC_OK) return st;
+            for (int64_t t = 0; t < gn; t++)
+                memcpy(rows.data() + (size_t)kv.second[t] * g.total * S,
+                       grp.data() + (size_t)t * g.total * S,
+                       (size_t)g.total * S);
+        }
         /* hashes for the healed shards, then assemble output streams */
         std::vector<uint8_t> sums((size_t)n * g.total * hsz);
-        st = mec_bitrot_sum_batch(ctx_, algo, (int)(n * g.total), rows.data(),
-                                  S, S, sums.data());
+        mec_status st = mec_bitrot_sum_batch(ctx_, algo, (int)(n * g.total),
+                                             rows.data(), S, S, sums.data());
         if (st != MEC_OK) return st;
         for (int s = 0; s < g.total; s++) {
             if (!out_bufs || !out_bufs[s]) continue;

@@ -347,3 +347,64 @@ def test_encode_parity_large_generic_geometries():
         # originally tripped: a NULL ctx reached the C-ABI)
         with pytest.raises(minio_amd.MecError):
             e.decode_data_and_parity_blocks(damaged)
+
+
+def test_heal_with_corrupt_drive():
+    # Heal reads through bitrot readers: a corrupt shard counts as missing
+    # (cmd/erasure-decode.go:322 + bitrot-streaming.go:185-197); with >= d
+    # intact shards per block the heal output is still exact
+    d, p, bs = 4, 2, 4096
+    total_len = 3 * bs + 777
+    data = rnd(total_len, SEED + 21)
+    with minio_amd.Erasure(d, p, bs) as e:
+        streams, _ = e.encode_stream(data)
+        S = e.shard_size()
+        dmg = list(streams)
+        dmg[0] = None  # missing drive to heal
+        corrupt = bytearray(streams[2])
+        corrupt[(32 + S) * 1 + 32 + 3] ^= 0xFF  # block 1 shard of drive 2
+        dmg[2] = bytes(corrupt)
+        healed = e.heal_stream(dmg, total_len)
+        assert healed[0] == streams[0]  # regenerated exactly
+        # too much loss in one block -> errFileCorrupt
+        dmg2 = list(streams)
+        dmg2[0] = None
+        dmg2[1] = None
+        c2 = bytearray(streams[2])
+        c2[32 + 3] ^= 1  # block 0 of drive 2 -> only 3 intact < d
+        dmg2[2] = bytes(c2)
+        with pytest.raises(minio_amd.MecError):
+            e.heal_stream(dmg2, total_len)
+
+
+def test_decode_random_offsets_fuzz():
+    # mirrors the random-offset fuzz at cmd/erasure-decode_test.go:200
+    d, p, bs = 8, 4, 64 * 1024
+    total_len = 7 * bs + 31415
+    data = rnd(total_len, SEED + 22)
+    rng = random.Random(99)
+    with minio_amd.Erasure(d, p, bs) as e:
+        streams, _ = e.encode_stream(data)
+        for _ in range(40):
+            off = rng.randrange(total_len)
+            ln = rng.randrange(1, total_len - off + 1)
+            dmg = list(streams)
+            for s in rng.sample(range(d + p), rng.randrange(0, p + 1)):
+                dmg[s] = None
+            assert e.decode_stream(dmg, total_len, off, ln) == \
+                data[off:off + ln], f"off={off} len={ln}"
+
+
+def test_fullsize_roundtrip_batch():
+    # full BASELINE shard sizes, encode -> erase p -> reconstruct, batch 16
+    for d, p, bs in [(8, 4, 1 << 20), (16, 4, 4 << 20)]:
+        n = 16
+        data = rnd(n * bs, SEED + 23 + d)
+        with minio_amd.Erasure(d, p, bs) as e:
+            shards, sums = e.encode_batch(data, bs, n, minio_amd.HIGHWAYHASH256S)
+            for b in (0, n - 1):
+                damaged = list(shards[b])
+                for i in range(p):
+                    damaged[d - 1 - i] = None  # erase p rows spanning data
+                rec = e.decode_data_and_parity_blocks(damaged)
+                assert rec == shards[b]
