@@ -209,7 +209,6 @@ class Erasure:
         sums = per-shard bitrot digests when algo given, else None."""
         if len(data) == 0:
             return [b""] * (self.d + self.p), None
-        s = shard_size(len(data), self.d)  # ceil(len/d)
         shards, sums = self.encode_batch(data, len(data), 1, algo)
         return ([sh for sh in shards[0]],
                 sums[0] if sums is not None else None)

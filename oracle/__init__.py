@@ -169,7 +169,7 @@ def encode_stream(d: int, p: int, block_size: int, data: bytes, algo: int):
     whole-file algorithms return (streams, whole_sums))."""
     rs = RS(d, p)
     streams = [b""] * (d + p)
-    whole = [bitrot_sum_stream_init(algo) for _ in range(d + p)] if algo != HIGHWAYHASH256S else None
+    whole = [[] for _ in range(d + p)] if algo != HIGHWAYHASH256S else None
     for off in range(0, max(len(data), 1), block_size):
         block = data[off:off + block_size]
         if not block and off > 0:
@@ -184,7 +184,3 @@ def encode_stream(d: int, p: int, block_size: int, data: bytes, algo: int):
     if algo == HIGHWAYHASH256S:
         return streams, None
     return streams, [bitrot_sum(algo, b"".join(w)) for w in whole]
-
-
-def bitrot_sum_stream_init(algo):
-    return []
