@@ -286,7 +286,7 @@ __device__ __forceinline__ void hh2_update(HH2 &s, uint64_t w0, uint64_t w1) {
  * ~340 cyc/packet vs ~104 issue-bound).  Interleaving NC independent
  * chains' packet updates in one lane fills those stalls. */
 template <int NC, bool RAGGED>
-__global__ void __launch_bounds__(256) hh256_batch_kernel(HashArgs a) {
+__global__ void __launch_bounds__(512) hh256_batch_kernel(HashArgs a) {
     const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     const int64_t slot = tid >> 1;
     const int h = (int)(tid & 1); /* 0: HH lanes {0,1}; 1: HH lanes {2,3} */
@@ -817,15 +817,20 @@ hipError_t mec_launch_hash(int algo, const HashArgs *args,
         /* NC=2 measured 2.3x slower even with the tail compiled out (the
          * doubled stream reaches 256 VGPR and the interleave does not
          * cover HH's dependency chains the way it does SHA's) — NC=1 for
-         * both cases; aligned lengths still skip the tail code. */
-        if (args->msg_len % 32 == 0) {
-            grid.x = (uint32_t)((args->n_chains * 2 + 255) / 256);
-            hipLaunchKernelGGL((hh256_batch_kernel<1, false>), grid, blk, 0,
-                               stream, *args);
-        } else {
-            grid.x = (uint32_t)((args->n_chains * 2 + 255) / 256);
-            hipLaunchKernelGGL((hh256_batch_kernel<1, true>), grid, blk, 0,
-                               stream, *args);
+         * both cases; aligned lengths still skip the tail code.
+         * Workgroup size via MEC_HH_WG: 512 packs 2 hash waves per SIMD
+         * (a wave's dependency stalls are covered by its co-resident
+         * partner); 256 spreads 1 wave/SIMD. */
+        {
+            static const int wg = gf_env_int("MEC_HH_WG", 256);
+            dim3 hblk((uint32_t)(wg >= 512 ? 512 : 256));
+            grid.x = (uint32_t)((args->n_chains * 2 + hblk.x - 1) / hblk.x);
+            if (args->msg_len % 32 == 0)
+                hipLaunchKernelGGL((hh256_batch_kernel<1, false>), grid,
+                                   hblk, 0, stream, *args);
+            else
+                hipLaunchKernelGGL((hh256_batch_kernel<1, true>), grid, hblk,
+                                   0, stream, *args);
         }
         break;
     case 4: /* BLAKE2b512 */
