@@ -58,8 +58,17 @@ __device__ __forceinline__ void xor4(uint4 &a, const uint4 &b) {
 /* W = uint4 columns per thread (16 or 32 B); NT = nontemporal parity
  * stores (parity is written once, never re-read by this kernel).
  * Variant selection via MEC_GF_W / MEC_GF_NT env (perf sweeps). */
-template <int D, int P, const uint8_t (&MAT)[P][D], int W, bool NT>
+template <int D, int P, const uint8_t (&MAT)[P][D], int W, bool NT,
+          int LDSPAD = 0>
 __global__ void __launch_bounds__(256) gf_encode_kernel(GfEncArgs a) {
+    /* LDSPAD > 0 caps workgroups/CU (160 KiB / LDSPAD) so co-resident
+     * hash waves of the pipelined previous batch keep issue share; the
+     * pad must be touched or it is elided */
+    if (LDSPAD > 0) {
+        __shared__ uint8_t pad[LDSPAD > 0 ? LDSPAD : 1];
+        if (threadIdx.x == 0xFFFFFFFF) pad[0] = 1; /* never true; keeps pad */
+        (void)pad;
+    }
     const int b = blockIdx.y;
     const int64_t cols = (a.shard_len + 16 * W - 1) / (16 * W);
     const uint8_t *__restrict__ sbase = a.data + (int64_t)b * D * a.row_stride;
@@ -729,10 +738,8 @@ hipError_t mec_launch_gf_encode_spec(int d, int p, const GfEncArgs *args,
     static const int env_w = gf_env_int("MEC_GF_W", 1);
     static const int env_nt = gf_env_int("MEC_GF_NT", 1);
     static const int env_wgx = gf_env_int("MEC_GF_WGX", 4);
-    /* dummy dynamic-LDS reservation caps GF occupancy per CU so the
-     * latency-critical hash waves of the pipelined previous batch keep
-     * issue slots (0 = uncapped) */
-    static const int env_lds = gf_env_int("MEC_GF_LDS", 0);
+    static const int env_cap = gf_env_int("MEC_GF_CAP", 0); /* wg/CU cap */
+    static const int env_lds = 0; /* dynamic-LDS cap was a no-op; see CAP */
     const int W = (env_w == 1) ? 1 : 2;
     const int64_t cols = (args->shard_len + 16 * W - 1) / (16 * W);
     int64_t max_x = (cols + 255) / 256;
@@ -743,7 +750,15 @@ hipError_t mec_launch_gf_encode_spec(int d, int p, const GfEncArgs *args,
     dim3 blk(256);
 #define X(D, P)                                                              \
     if (d == D && p == P) {                                                  \
-        if (W == 1 && !env_nt)                                               \
+        if (env_cap == 4 && W == 1 && env_nt)                                \
+            hipLaunchKernelGGL(                                              \
+                (gf_encode_kernel<D, P, MAT_##D##_##P, 1, true, 36864>),     \
+                grid, blk, env_lds, stream, *args);                          \
+        else if (env_cap == 2 && W == 1 && env_nt)                           \
+            hipLaunchKernelGGL(                                              \
+                (gf_encode_kernel<D, P, MAT_##D##_##P, 1, true, 65536>),     \
+                grid, blk, env_lds, stream, *args);                          \
+        else if (W == 1 && !env_nt)                                          \
             hipLaunchKernelGGL((gf_encode_kernel<D, P, MAT_##D##_##P, 1,     \
                                                  false>),                    \
                                grid, blk, env_lds, stream, *args);           \
