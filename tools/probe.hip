@@ -91,6 +91,34 @@ __global__ void __launch_bounds__(256) hh_reg_probe(uint64_t *sink,
     sink[tid] = s.v0[0] ^ s.v1[1] ^ s.mul0[0] ^ s.mul1[1];
 }
 
+/* 2 independent chains per lane, register-only: if per-chain cost drops
+ * toward the issue floor, the product NC=2 failures were memory/VGPR-side
+ * and a leaner NC=2 is worth building; if not, the chain stalls are not
+ * coverable by lane-local ILP. */
+__global__ void __launch_bounds__(256) hh_reg_probe2(uint64_t *sink,
+                                                     int packets) {
+    const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    HH2 s[2];
+#pragma unroll
+    for (int u = 0; u < 2; u++)
+#pragma unroll
+        for (int j = 0; j < 2; j++) {
+            s[u].v0[j] = 0x1111111111111111ull * (j + 1) + tid + u;
+            s[u].v1[j] = 0x2222222222222222ull * (j + 1) ^ (tid - u);
+            s[u].mul0[j] = 0x3333333333333333ull * (j + 1);
+            s[u].mul1[j] = 0x4444444444444444ull * (j + 1);
+        }
+    uint64_t w0 = tid, w1 = ~tid;
+#pragma unroll 8
+    for (int t = 0; t < packets; t++) {
+        hh2_update(s[0], w0, w1);
+        hh2_update(s[1], w1, w0);
+        w0 += 0x9e3779b97f4a7c15ull;
+        w1 ^= w0;
+    }
+    sink[tid] = s[0].v0[0] ^ s[0].v1[1] ^ s[1].mul0[0] ^ s[1].mul1[1];
+}
+
 __global__ void __launch_bounds__(256) gf_mem_probe(const uint8_t *data,
                                                     uint8_t *parity,
                                                     int64_t stride,
@@ -143,6 +171,19 @@ int main() {
     CK(hipEventElapsedTime(&ms, e0, e1));
     printf("hh_reg_probe (no memory): %.3f ms/launch (real hash kernel "
            "~0.47)\n", ms / 5);
+
+    /* ---- hh_reg2: 12288 lanes x 2 chains x 4096 packets (same total) */
+    dim3 g1b(12288 / 256);
+    hipLaunchKernelGGL(hh_reg_probe2, g1b, b1, 0, 0, sink, 4096);
+    CK(hipDeviceSynchronize());
+    CK(hipEventRecord(e0));
+    for (int r = 0; r < 5; r++)
+        hipLaunchKernelGGL(hh_reg_probe2, g1b, b1, 0, 0, sink, 4096);
+    CK(hipEventRecord(e1));
+    CK(hipEventSynchronize(e1));
+    CK(hipEventElapsedTime(&ms, e0, e1));
+    printf("hh_reg_probe2 (2 chains/lane, half lanes): %.3f ms/launch\n",
+           ms / 5);
 
     /* ---- gf_mem: the 12-stream layout with no ladder */
     uint8_t *data, *par;
