@@ -84,7 +84,8 @@ struct mec_ctx {
 
     /* grow-only scratch (device + pinned host) for host-pointer calls */
     void *dev_a = nullptr, *dev_b = nullptr, *dev_c = nullptr;
-    size_t cap_a = 0, cap_b = 0, cap_c = 0;
+    void *dev_d = nullptr; /* stream-assembly output */
+    size_t cap_a = 0, cap_b = 0, cap_c = 0, cap_d = 0;
     void *pin = nullptr;
     size_t cap_pin = 0;
 
@@ -217,6 +218,7 @@ void mec_ctx_destroy(mec_ctx *ctx) {
     if (ctx->dev_a) (void)hipFree(ctx->dev_a);
     if (ctx->dev_b) (void)hipFree(ctx->dev_b);
     if (ctx->dev_c) (void)hipFree(ctx->dev_c);
+    if (ctx->dev_d) (void)hipFree(ctx->dev_d);
     if (ctx->pin) (void)hipHostFree(ctx->pin);
     if (ctx->ev_start) (void)hipEventDestroy(ctx->ev_start);
     if (ctx->ev_stop) (void)hipEventDestroy(ctx->ev_stop);
@@ -718,6 +720,74 @@ mec_status mec_timer_start(mec_ctx *ctx) {
     if (!ctx) return MEC_ERR_INVALID_ARG;
     HIP_TRY(hipSetDevice(ctx->device));
     HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
+    return MEC_OK;
+}
+
+/* GPU-side streaming-format assembly for the full-block phase of
+ * mec_encode_stream: upload packed object bytes once,
+ * scatter to shard rows on device, fused encode+hash, interleave the
+ * per-drive [hash||shard]* streams on device, copy one stream per drive
+ * out.  Layout matches cmd/bitrot-streaming.go:57-75 bit-for-bit (pinned
+ * by the stream round-trip tests). */
+mec_status mec_encode_stream_gpu(mec_ctx *ctx, const uint8_t *src,
+                                 int64_t n_full, int algo,
+                                 uint8_t *const *drive_bufs,
+                                 int64_t drive_off) {
+    if (!ctx) return MEC_ERR_INVALID_ARG;
+    std::lock_guard<std::mutex> lk(ctx->mu);
+    HIP_TRY(hipSetDevice(ctx->device));
+    const int d = ctx->d, p = ctx->p, total = d + p;
+    const int64_t bs = ctx->block_size, S = ctx->S, stride = ctx->stride;
+    const int64_t pitch = 32 + S;
+    const size_t src_bytes = (size_t)n_full * bs;
+    const size_t row_bytes = (size_t)n_full * d * stride;
+    const size_t par_bytes = (size_t)n_full * p * stride;
+    const size_t sum_bytes = (size_t)n_full * total * 32;
+    const size_t out_bytes = (size_t)total * n_full * pitch;
+    mec_status st;
+    if ((st = ctx->ensure(&ctx->dev_a, &ctx->cap_a,
+                          row_bytes > src_bytes ? row_bytes : src_bytes)) !=
+        MEC_OK)
+        return st;
+    if ((st = ctx->ensure(&ctx->dev_b, &ctx->cap_b, par_bytes)) != MEC_OK)
+        return st;
+    if ((st = ctx->ensure(&ctx->dev_c, &ctx->cap_c, sum_bytes)) != MEC_OK)
+        return st;
+    if ((st = ctx->ensure(&ctx->dev_d, &ctx->cap_d,
+                          out_bytes + row_bytes)) != MEC_OK)
+        return st;
+    /* dev_d holds [stream out | shard rows]; dev_a stages the packed src */
+    uint8_t *dev_rows = (uint8_t *)ctx->dev_d + out_bytes;
+    HIP_TRY(hipMemcpyAsync(ctx->dev_a, src, src_bytes, hipMemcpyHostToDevice,
+                           ctx->stream));
+    ScatterArgs sa{};
+    sa.src = (const uint8_t *)ctx->dev_a;
+    sa.rows = dev_rows;
+    sa.block_len = bs;
+    sa.S = S;
+    sa.row_stride = stride;
+    sa.n = n_full;
+    sa.d = d;
+    HIP_TRY(mec_launch_scatter_rows(&sa, ctx->stream));
+    st = encode_dev_locked(ctx, (int)n_full, dev_rows, bs, ctx->dev_b, algo,
+                           ctx->dev_c);
+    if (st != MEC_OK) return st;
+    InterleaveArgs ia{};
+    ia.data = dev_rows;
+    ia.parity = (const uint8_t *)ctx->dev_b;
+    ia.sums = (const uint8_t *)ctx->dev_c;
+    ia.out = (uint8_t *)ctx->dev_d;
+    ia.S = S;
+    ia.row_stride = stride;
+    ia.n = n_full;
+    ia.d = d;
+    ia.p = p;
+    HIP_TRY(mec_launch_stream_interleave(&ia, ctx->stream));
+    HIP_TRY(hipStreamSynchronize(ctx->stream));
+    for (int s = 0; s < total; s++)
+        HIP_TRY(hipMemcpy(drive_bufs[s] + drive_off,
+                          (uint8_t *)ctx->dev_d + (int64_t)s * n_full * pitch,
+                          (size_t)(n_full * pitch), hipMemcpyDeviceToHost));
     return MEC_OK;
 }
 

@@ -67,10 +67,13 @@ int64_t stream_off(const Geo &g, int64_t b, int algo) {
 
 } // namespace
 
-/* geometry getters (defined in ec_abi.cpp) */
+/* geometry getters + GPU stream assembly (defined in ec_abi.cpp) */
 extern "C" int64_t mec_ctx_block_size(mec_ctx *);
 extern "C" int mec_ctx_d(mec_ctx *);
 extern "C" int mec_ctx_p(mec_ctx *);
+extern "C" mec_status mec_encode_stream_gpu(mec_ctx *, const uint8_t *,
+                                            int64_t, int, uint8_t *const *,
+                                            int64_t);
 
 extern "C" {
 
@@ -103,9 +106,20 @@ mec_status mec_encode_stream(mec_ctx *ctx_, const uint8_t *src,
     /* encode all full blocks in one batch, the ragged last block alone */
     int64_t n_full = g.n_blocks - (g.last_len != g.block_size ? 1 : 0);
     std::vector<uint8_t> parity, sums;
+    /* full blocks of the streaming format: fully on-device assembly
+     * (scatter -> fused encode+hash -> [hash||shard]* interleave), one
+     * upload + one download per drive */
+    bool gpu_full_done = false;
+    if (streaming && n_full > 0) {
+        mec_status st = mec_encode_stream_gpu(ctx_, src, n_full, algo,
+                                              drive_bufs, 0);
+        if (st != MEC_OK) return st;
+        gpu_full_done = true;
+    }
     /* whole-file accumulation needs every shard byte per drive; we assemble
      * drive streams first, then hash them in one strided batch call */
     for (int64_t phase = 0; phase < 2; phase++) {
+        if (phase == 0 && gpu_full_done) continue;
         int64_t b0 = phase == 0 ? 0 : n_full;
         int64_t nb = phase == 0 ? n_full : g.n_blocks - n_full;
         if (nb <= 0) continue;

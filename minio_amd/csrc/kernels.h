@@ -63,7 +63,32 @@ struct FusedArgs {
     uint64_t key[4];
 };
 
+struct ScatterArgs {
+    const uint8_t *src;
+    uint8_t *rows;
+    int64_t block_len;
+    int64_t S;
+    int64_t row_stride;
+    int64_t n;
+    int d;
+};
+
+struct InterleaveArgs {
+    const uint8_t *data;
+    const uint8_t *parity;
+    const uint8_t *sums;
+    uint8_t *out;
+    int64_t S;
+    int64_t row_stride;
+    int64_t n;
+    int d, p;
+};
+
 extern "C" {
+hipError_t mec_launch_scatter_rows(const ScatterArgs *args,
+                                   hipStream_t stream);
+hipError_t mec_launch_stream_interleave(const InterleaveArgs *args,
+                                        hipStream_t stream);
 hipError_t mec_launch_fused_encode_hh(int d, int p, const FusedArgs *args,
                                       hipStream_t stream);
 hipError_t mec_launch_fused2_encode_hh(int d, int p, const FusedArgs *args,
