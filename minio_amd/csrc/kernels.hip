@@ -750,12 +750,12 @@ __global__ void __launch_bounds__(256) scatter_rows_kernel(ScatterArgs a) {
             if (((uintptr_t)sp & 3) == 0) {
                 v.x = p4[0]; v.y = p4[1]; v.z = p4[2]; v.w = p4[3];
             } else {
-                uint8_t tmp[16];
+                alignas(16) uint8_t tmp[16];
                 for (int i = 0; i < 16; i++) tmp[i] = sp[i];
                 v = *(uint4 *)tmp;
             }
         } else if (src_off < a.block_len) {
-            uint8_t tmp[16] = {0};
+            alignas(16) uint8_t tmp[16] = {0};
             const int have = (int)(a.block_len - src_off);
             for (int i = 0; i < have; i++) tmp[i] = sp[i];
             v = *(uint4 *)tmp;
@@ -781,9 +781,14 @@ __global__ void __launch_bounds__(256) stream_interleave_kernel(
         const int64_t o = (r % per_entry) * 16;
         uint8_t *dst = a.out + ((int64_t)s * a.n + b) * pitch + o;
         if (o < 32) {
-            /* hash half: two 16-B units */
+            /* hash half: two 16-B units; dst is 16-B aligned only when
+             * pitch%16==0 (ragged S, e.g. EC12+4, makes entry bases odd) */
             const uint8_t *h = a.sums + (b * total + s) * 32 + o;
-            *(uint4 *)dst = *(const uint4 *)h;
+            if ((((int64_t)s * a.n + b) * pitch + o) % 16 == 0) {
+                *(uint4 *)dst = *(const uint4 *)h;
+            } else {
+                for (int i = 0; i < 16; i++) dst[i] = h[i];
+            }
         } else {
             const int64_t j = o - 32;
             const uint8_t *row = (s < a.d)
