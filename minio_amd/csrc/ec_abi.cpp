@@ -190,8 +190,46 @@ mec_status mec_ctx_create(int d, int p, int64_t block_size, int device,
         return MEC_ERR_INVALID_ARG;
     }
     hipError_t e = hipSetDevice(device);
-    if (e == hipSuccess) e = hipStreamCreate(&ctx->stream);
-    if (e == hipSuccess) e = hipStreamCreate(&ctx->stream2);
+    /* MEC_CU_SPLIT=H: partition the chip so the latency-bound hash lane
+     * (stream2) gets H CUs EXCLUSIVELY and the memory-bound GF lane
+     * (stream) the rest.  Overlapped gf/hash waves sharing a SIMD stretch
+     * the hash ~1.5x (issue-slot contention); an exclusive partition
+     * removes that.  CUs are spread round-robin so both lanes touch every
+     * XCD's L2/HBM channels.  0/unset = classic shared streams. */
+    int cu_split = 0;
+    if (const char *s = getenv("MEC_CU_SPLIT")) cu_split = atoi(s);
+    if (cu_split > 0) {
+        int ncu = 0;
+        (void)hipDeviceGetAttribute(&ncu,
+                                    hipDeviceAttributeMultiprocessorCount,
+                                    device);
+        if (ncu <= 0 || cu_split >= ncu) cu_split = 0;
+        if (cu_split > 0) {
+            uint32_t mh[8] = {0}, mg[8] = {0};
+            int given = 0;
+            for (int i = 0; i < ncu && i < 256; i++) {
+                /* spread hash CUs evenly across the index space (and so
+                 * across XCDs, which interleave in the physical mapping) */
+                bool h = ((int64_t)(i + 1) * cu_split / ncu) >
+                         ((int64_t)i * cu_split / ncu);
+                if (h) {
+                    mh[i >> 5] |= 1u << (i & 31);
+                    given++;
+                } else {
+                    mg[i >> 5] |= 1u << (i & 31);
+                }
+            }
+            (void)given;
+            if (e == hipSuccess)
+                e = hipExtStreamCreateWithCUMask(&ctx->stream, 8, mg);
+            if (e == hipSuccess)
+                e = hipExtStreamCreateWithCUMask(&ctx->stream2, 8, mh);
+        }
+    }
+    if (cu_split <= 0) {
+        if (e == hipSuccess) e = hipStreamCreate(&ctx->stream);
+        if (e == hipSuccess) e = hipStreamCreate(&ctx->stream2);
+    }
     if (e == hipSuccess) e = hipEventCreate(&ctx->ev_start);
     if (e == hipSuccess) e = hipEventCreate(&ctx->ev_stop);
     if (e == hipSuccess) e = hipEventCreateWithFlags(&ctx->ev_gf, hipEventDisableTiming);

@@ -35,8 +35,10 @@
 namespace fused {
 
 __device__ __forceinline__ uint32_t gf2x(uint32_t x) {
-    uint32_t hi = x & 0x80808080u;
-    return ((x & 0x7f7f7f7fu) << 1) ^ ((hi >> 7) * 0x1du);
+    /* full-rate v_perm reduction select; see kernels.hip gf2x */
+    uint32_t sel = (x & 0x80808080u) >> 7;
+    return ((x << 1) & 0xfefefefeu) ^
+           __builtin_amdgcn_perm(0u, 0x00001d00u, sel);
 }
 
 __device__ __forceinline__ void gf2x4(uint4 &v) {

@@ -38,8 +38,10 @@
 namespace fused2 {
 
 __device__ __forceinline__ uint32_t gf2x(uint32_t x) {
-    uint32_t hi = x & 0x80808080u;
-    return ((x & 0x7f7f7f7fu) << 1) ^ ((hi >> 7) * 0x1du);
+    /* full-rate v_perm reduction select; see kernels.hip gf2x */
+    uint32_t sel = (x & 0x80808080u) >> 7;
+    return ((x << 1) & 0xfefefefeu) ^
+           __builtin_amdgcn_perm(0u, 0x00001d00u, sel);
 }
 
 __device__ __forceinline__ void gf2x4(uint4 &v) {
@@ -183,8 +185,10 @@ __global__ void __launch_bounds__(512) fused2_encode_hh_kernel(FusedArgs a) {
                                    off));
                 }
             }
-            /* this wave's LDS writes must land before the publish */
-            __builtin_amdgcn_s_waitcnt(0); /* vmcnt(0) lgkmcnt(0) */
+            /* this wave's LDS (ds_write) traffic must land before the
+             * publish; global parity stores are not part of the handoff,
+             * so leave vmcnt unconstrained (lgkmcnt(0) only) */
+            __builtin_amdgcn_s_waitcnt(0xc07f); /* lgkmcnt(0) */
             if ((tid & 63) == 0) {
                 int prev = __hip_atomic_fetch_add(&flags[4 + slot], 1,
                                                   __ATOMIC_RELAXED,
@@ -234,7 +238,7 @@ __global__ void __launch_bounds__(512) fused2_encode_hh_kernel(FusedArgs a) {
         if (act) {
             /* chains are the latency wall: outprioritize the producer
              * waves sharing this SIMD (T5) */
-            __builtin_amdgcn_s_setprio(1);
+            __builtin_amdgcn_s_setprio(3);
             const uint8_t *row =
                 &lds[slot * SLOT + (cg * TOT + cs) * ROW + 16 * h];
 #pragma unroll 4
@@ -245,7 +249,7 @@ __global__ void __launch_bounds__(512) fused2_encode_hh_kernel(FusedArgs a) {
             }
             __builtin_amdgcn_s_setprio(0);
         }
-        __builtin_amdgcn_s_waitcnt(0);
+        __builtin_amdgcn_s_waitcnt(0xc07f); /* lgkmcnt(0): ds_reads done */
         if ((tid & 63) == 0) {
             int prev = __hip_atomic_fetch_add(&flags[6 + slot], 1,
                                               __ATOMIC_RELAXED,

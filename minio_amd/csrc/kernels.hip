@@ -26,10 +26,24 @@
 
 /* ---- GF(2^8) SWAR helpers --------------------------------------------- */
 
+/* gfx950 3-input boolean LUT op (1 full-rate VALU slot) */
+__device__ __forceinline__ uint32_t xor3(uint32_t a, uint32_t b, uint32_t c) {
+    return __builtin_amdgcn_bitop3_b32(a, b, c, 0x96); /* a^b^c */
+}
+__device__ __forceinline__ uint32_t andxor(uint32_t a, uint32_t b,
+                                           uint32_t c) {
+    return __builtin_amdgcn_bitop3_b32(a, b, c, 0x6a); /* (a&b)^c */
+}
+
 __device__ __forceinline__ uint32_t gf2x(uint32_t x) {
-    /* multiply 4 packed GF(2^8) bytes by 2 (poly 0x11D) */
-    uint32_t hi = x & 0x80808080u;
-    return ((x & 0x7f7f7f7fu) << 1) ^ ((hi >> 7) * 0x1du);
+    /* multiply 4 packed GF(2^8) bytes by 2 (poly 0x11D) in 5 full-rate
+     * slots: the reduction byte (0x00 or 0x1d per sign bit) comes from a
+     * v_perm_b32 byte-select instead of a multiply, and the shifted-out
+     * cross-byte bits are cleared by the same bitop3 that applies the
+     * reduction ((a&b)^c). */
+    uint32_t sel = (x & 0x80808080u) >> 7;
+    uint32_t m = __builtin_amdgcn_perm(0u, 0x00001d00u, sel);
+    return andxor(x << 1, 0xfefefefeu, m);
 }
 
 __device__ __forceinline__ void gf2x4(uint4 &v) {
@@ -44,6 +58,14 @@ __device__ __forceinline__ void xor4(uint4 &a, const uint4 &b) {
     a.y ^= b.y;
     a.z ^= b.z;
     a.w ^= b.w;
+}
+
+__device__ __forceinline__ void xor34(uint4 &a, const uint4 &b,
+                                      const uint4 &c) {
+    a.x = xor3(a.x, b.x, c.x);
+    a.y = xor3(a.y, b.y, c.y);
+    a.z = xor3(a.z, b.z, c.z);
+    a.w = xor3(a.w, b.w, c.w);
 }
 
 /* ---- specialized encode kernels (constexpr matrix) ---------------------
@@ -95,24 +117,50 @@ __global__ void __launch_bounds__(256) gf_encode_kernel(GfEncArgs a) {
         }
 #pragma unroll
         for (int k = 0; k < D; k++) {
-            uint4 pw[W];
+            /* xtime chain walked two bits at a time so each parity's
+             * accumulate can fold bit-pairs into one v_bitop3 xor3; all
+             * bit tests are constexpr on MAT and fold at compile time */
+            uint4 cur[W], nxt[W];
 #pragma unroll
-            for (int w = 0; w < W; w++) pw[w] = pws[k][w];
+            for (int w = 0; w < W; w++) cur[w] = pws[k][w];
 #pragma unroll
-            for (int bit = 0; bit < 8; bit++) {
-                uint32_t need = 0;
+            for (int bit = 0; bit < 8; bit += 2) {
+                uint32_t needCur = 0, needHi = 0;
 #pragma unroll
-                for (int i = 0; i < P; i++)
-                    need |= (uint32_t)MAT[i][k] >> bit;
-                if (!need) break; /* compile-time folded */
-                if (bit)
+                for (int i = 0; i < P; i++) {
+                    needCur |= (uint32_t)MAT[i][k] >> bit;
+                    needHi |= (uint32_t)MAT[i][k] >> (bit + 1);
+                }
+                if (!needCur) break; /* compile-time folded */
+                if (needHi)
 #pragma unroll
-                    for (int w = 0; w < W; w++) gf2x4(pw[w]);
+                    for (int w = 0; w < W; w++) {
+                        nxt[w] = cur[w];
+                        gf2x4(nxt[w]);
+                    }
 #pragma unroll
-                for (int i = 0; i < P; i++)
-                    if ((MAT[i][k] >> bit) & 1)
+                for (int i = 0; i < P; i++) {
+                    const int b0 = (MAT[i][k] >> bit) & 1;
+                    const int b1 = (MAT[i][k] >> (bit + 1)) & 1;
+                    if (b0 && b1)
 #pragma unroll
-                        for (int w = 0; w < W; w++) xor4(acc[i][w], pw[w]);
+                        for (int w = 0; w < W; w++)
+                            xor34(acc[i][w], cur[w], nxt[w]);
+                    else if (b0)
+#pragma unroll
+                        for (int w = 0; w < W; w++) xor4(acc[i][w], cur[w]);
+                    else if (b1)
+#pragma unroll
+                        for (int w = 0; w < W; w++) xor4(acc[i][w], nxt[w]);
+                }
+                if (needHi >> 1)
+#pragma unroll
+                    for (int w = 0; w < W; w++) {
+                        cur[w] = nxt[w];
+                        gf2x4(cur[w]);
+                    }
+                else
+                    break; /* compile-time folded */
             }
         }
 #pragma unroll
@@ -456,6 +504,164 @@ __global__ void __launch_bounds__(512) hh256_batch_kernel(HashArgs a) {
         out.z = (uint32_t)o1;
         out.w = (uint32_t)(o1 >> 32);
         *(uint4 *)(a.sums + sum_idx[u] * 32 + 16 * h) = out;
+    }
+}
+
+/* ---- LDS-staged HighwayHash (MEC_HH_LDS=1) -----------------------------
+ *
+ * Same chain math as hh256_batch_kernel (pair-lanes, v_perm zipper), but
+ * the packet stream is staged through LDS with direct global->LDS loads
+ * (global_load_lds_dwordx4) instead of per-lane strided VGPR prefetch:
+ *  - each wave stages ITS OWN 32 chains' next 512-B tiles (wave-private:
+ *    no barriers, only counted vmcnt), so HBM sees 64 long coalesced
+ *    512-B bursts per tile instead of 32-B scatters from 24k lanes;
+ *  - the 128-KiB static LDS footprint forces 1 WG/CU, and leaves <32 KiB
+ *    free, so a co-scheduled GF kernel built with the MEC_GF_CAP=4 pad
+ *    (36 KiB) CANNOT share the CU: hash waves get exclusive SIMDs during
+ *    pipelined encode without CU masks (which gfx950 ignores).
+ */
+__device__ __forceinline__ void hh_lds_stage16(const uint8_t *const *src,
+                                               int64_t off, uint8_t *lds) {
+#pragma unroll
+    for (int k = 0; k < 16; k++)
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) uint32_t
+                 *)(src[k] + off),
+            (__attribute__((address_space(3))) uint32_t
+                 *)(lds + (int64_t)k * 1024),
+            16, 0, 0);
+}
+
+template <bool RAGGED>
+__global__ void __launch_bounds__(256) hh256_lds_kernel(HashArgs a) {
+    constexpr int T = 512;   /* bytes per chain per tile */
+    constexpr int CPW = 32;  /* chains per wave (pair-lanes) */
+    __shared__ uint8_t stage[4][2][CPW * T]; /* 128 KiB, wave-private */
+
+    const int wv = (int)(threadIdx.x >> 6), lane = (int)(threadIdx.x & 63);
+    const int64_t wave_c0 = ((int64_t)blockIdx.x * 4 + wv) * CPW;
+    const int rc = lane >> 1;
+    const int h = lane & 1;
+    const int64_t chain = wave_c0 + rc;
+    const bool act = chain < a.n_chains;
+    int64_t sum_idx;
+    const uint8_t *msg = chain_ptr(a, act ? chain : 0, sum_idx);
+
+    /* per-lane glds sources: glds k stages chains 2k (lanes 0-31) and
+     * 2k+1 (lanes 32-63), 16 B per lane = 1024 B contiguous LDS */
+    const uint8_t *src[CPW / 2];
+#pragma unroll
+    for (int k = 0; k < CPW / 2; k++) {
+        int64_t c = wave_c0 + 2 * k + (lane >> 5);
+        if (c >= a.n_chains) c = 0; /* clamp: data unused */
+        int64_t si;
+        src[k] = chain_ptr(a, c, si) + (int64_t)(lane & 31) * 16;
+    }
+
+    const uint64_t init0[4] = {0xdbe6d5d5fe4cce2full, 0xa4093822299f31d0ull,
+                               0x13198a2e03707344ull, 0x243f6a8885a308d3ull};
+    const uint64_t init1[4] = {0x3bd39e10cb0ef593ull, 0xc0acf169b5f18a8cull,
+                               0xbe5466cf34e90c6cull, 0x452821e638d01377ull};
+    HH2 s;
+#pragma unroll
+    for (int j = 0; j < 2; j++) {
+        int li = 2 * h + j;
+        s.mul0[j] = init0[li];
+        s.mul1[j] = init1[li];
+        s.v0[j] = init0[li] ^ a.key[li];
+        s.v1[j] = init1[li] ^ ((a.key[li] >> 32) | (a.key[li] << 32));
+    }
+
+    int64_t len = a.msg_len;
+    const int64_t ntiles = len / T;
+    __builtin_amdgcn_s_setprio(1);
+    if (ntiles > 0) {
+        hh_lds_stage16(src, 0, &stage[wv][0][0]);
+        for (int64_t it = 0; it < ntiles; it++) {
+            const int buf = (int)(it & 1);
+            if (it + 1 < ntiles)
+                hh_lds_stage16(src, (it + 1) * T, &stage[wv][buf ^ 1][0]);
+            /* wait for THIS buffer's 16 glds (the next tile's 16 may stay
+             * in flight): vmcnt(16), lgkm/exp unconstrained */
+            if (it + 1 < ntiles)
+                __builtin_amdgcn_s_waitcnt(0x4f70); /* vmcnt(16) */
+            else
+                __builtin_amdgcn_s_waitcnt(0x0f70); /* vmcnt(0) */
+            const uint8_t *row = &stage[wv][buf][rc * T + 16 * h];
+#pragma unroll
+            for (int t = 0; t < T / 32; t++) {
+                uint4 q = *(const uint4 *)(row + 32 * t);
+                hh2_update(s, (uint64_t)q.x | ((uint64_t)q.y << 32),
+                           (uint64_t)q.z | ((uint64_t)q.w << 32));
+            }
+        }
+        len -= ntiles * T;
+    }
+    /* sub-tile remainder straight from global (0..511 B) */
+    const uint8_t *mp = msg + ntiles * T + 16 * h;
+    while (len >= 32) {
+        uint4 q = *(const uint4 *)mp;
+        hh2_update(s, (uint64_t)q.x | ((uint64_t)q.y << 32),
+                   (uint64_t)q.z | ((uint64_t)q.w << 32));
+        mp += 32;
+        len -= 32;
+    }
+    __builtin_amdgcn_s_setprio(0);
+    if (RAGGED && len > 0) {
+        /* UpdateRemainder — identical to hh256_batch_kernel's tail */
+        const int mod32 = (int)len;
+        const int mod4 = mod32 & 3;
+        const uint8_t *tail_msg = mp - 16 * h;
+#pragma unroll
+        for (int j = 0; j < 2; j++) {
+            s.v0[j] += ((uint64_t)mod32 << 32) + (uint64_t)mod32;
+            uint32_t h0 = (uint32_t)s.v1[j];
+            uint32_t h1 = (uint32_t)(s.v1[j] >> 32);
+            s.v1[j] = (uint32_t)((h0 << mod32) | (h0 >> (32 - mod32)));
+            s.v1[j] |= (uint64_t)((h1 << mod32) | (h1 >> (32 - mod32)))
+                       << 32;
+        }
+        uint8_t packet[32];
+#pragma unroll
+        for (int i = 0; i < 32; i++) packet[i] = 0;
+        for (int i = 0; i < (mod32 & ~3); i++) packet[i] = tail_msg[i];
+        const uint8_t *rem = tail_msg + (mod32 & ~3);
+        if (mod32 & 16) {
+            for (int i = 0; i < 4; i++)
+                packet[28 + i] = rem[i + mod4 - 4];
+        } else if (mod4) {
+            packet[16] = rem[0];
+            packet[17] = rem[mod4 >> 1];
+            packet[18] = rem[mod4 - 1];
+        }
+        uint64_t w[2];
+#pragma unroll
+        for (int j = 0; j < 2; j++) {
+            uint64_t v = 0;
+            for (int bt = 7; bt >= 0; bt--)
+                v = (v << 8) | packet[16 * h + 8 * j + bt];
+            w[j] = v;
+        }
+        hh2_update(s, w[0], w[1]);
+    }
+#pragma unroll 1
+    for (int r = 0; r < 10; r++) {
+        uint64_t p0 = shfl_x(s.v0[0], 1);
+        uint64_t p1 = shfl_x(s.v0[1], 1);
+        hh2_update(s, (p0 >> 32) | (p0 << 32), (p1 >> 32) | (p1 << 32));
+    }
+    if (act) {
+        uint64_t a2 = s.v1[0] + s.mul1[0];
+        uint64_t a3 = (s.v1[1] + s.mul1[1]) & 0x3fffffffffffffffull;
+        uint64_t o0 = (s.v0[0] + s.mul0[0]) ^ (a2 << 1) ^ (a2 << 2);
+        uint64_t o1 = (s.v0[1] + s.mul0[1]) ^ ((a3 << 1) | (a2 >> 63)) ^
+                      ((a3 << 2) | (a2 >> 62));
+        uint4 out;
+        out.x = (uint32_t)o0;
+        out.y = (uint32_t)(o0 >> 32);
+        out.z = (uint32_t)o1;
+        out.w = (uint32_t)(o1 >> 32);
+        *(uint4 *)(a.sums + sum_idx * 32 + 16 * h) = out;
     }
 }
 
@@ -949,7 +1155,19 @@ hipError_t mec_launch_hash(int algo, const HashArgs *args,
          * (a wave's dependency stalls are covered by its co-resident
          * partner); 256 spreads 1 wave/SIMD. */
         {
+            static const int use_lds = gf_env_int("MEC_HH_LDS", 0);
             static const int wg = gf_env_int("MEC_HH_WG", 256);
+            if (use_lds && args->msg_len >= 512) {
+                dim3 hblk(256);
+                grid.x = (uint32_t)((args->n_chains + 127) / 128);
+                if (args->msg_len % 32 == 0)
+                    hipLaunchKernelGGL((hh256_lds_kernel<false>), grid,
+                                       hblk, 0, stream, *args);
+                else
+                    hipLaunchKernelGGL((hh256_lds_kernel<true>), grid, hblk,
+                                       0, stream, *args);
+                break;
+            }
             dim3 hblk((uint32_t)(wg >= 512 ? 512 : 256));
             grid.x = (uint32_t)((args->n_chains * 2 + hblk.x - 1) / hblk.x);
             if (args->msg_len % 32 == 0)
