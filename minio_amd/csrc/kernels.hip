@@ -703,16 +703,19 @@ __device__ void sha256_block(uint32_t h[8], const uint32_t w_in[16]) {
             wi = w[i];
         } else {
             uint32_t w15 = w[(i - 15) & 15], w2 = w[(i - 2) & 15];
-            uint32_t s0 = rotr32(w15, 7) ^ rotr32(w15, 18) ^ (w15 >> 3);
-            uint32_t s1 = rotr32(w2, 17) ^ rotr32(w2, 19) ^ (w2 >> 10);
+            uint32_t s0 = xor3(rotr32(w15, 7), rotr32(w15, 18), w15 >> 3);
+            uint32_t s1 = xor3(rotr32(w2, 17), rotr32(w2, 19), w2 >> 10);
             wi = w[i & 15] + s0 + w[(i - 7) & 15] + s1;
             w[i & 15] = wi;
         }
-        uint32_t S1 = rotr32(e, 6) ^ rotr32(e, 11) ^ rotr32(e, 25);
-        uint32_t ch = (e & f) ^ (~e & g);
+        /* Ch/Maj/Sigma as single v_bitop3_b32 LUT ops (gfx950): Ch and
+         * Maj sit ON the round's dependency chain, so this shortens the
+         * critical path as well as the issue count */
+        uint32_t S1 = xor3(rotr32(e, 6), rotr32(e, 11), rotr32(e, 25));
+        uint32_t ch = __builtin_amdgcn_bitop3_b32(e, f, g, 0xca);
         uint32_t t1 = hh + S1 + ch + SHA_K[i] + wi;
-        uint32_t S0 = rotr32(a, 2) ^ rotr32(a, 13) ^ rotr32(a, 22);
-        uint32_t maj = (a & b) ^ (a & c) ^ (b & c);
+        uint32_t S0 = xor3(rotr32(a, 2), rotr32(a, 13), rotr32(a, 22));
+        uint32_t maj = __builtin_amdgcn_bitop3_b32(a, b, c, 0xe8);
         uint32_t t2 = S0 + maj;
         hh = g; g = f; f = e; e = d + t1;
         d = c; c = b; b = a; a = t1 + t2;
@@ -724,8 +727,8 @@ __device__ void sha256_block(uint32_t h[8], const uint32_t w_in[16]) {
 /* NC independent chains per lane: SHA-256's round chain is strictly serial
  * (measured ~17 cyc/instr effective at 1 chain/lane, 1 wave/SIMD — pure
  * dependency latency); interleaving NC chains fills the stalls. */
-template <int NC>
-__global__ void __launch_bounds__(256) sha256_batch_kernel(HashArgs a) {
+template <int NC, int LB = 256>
+__global__ void __launch_bounds__(LB, 1) sha256_batch_kernel(HashArgs a) {
     const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     const int64_t c0 = tid * NC;
     if (c0 >= a.n_chains) return;
@@ -1135,12 +1138,28 @@ hipError_t mec_launch_hash(int algo, const HashArgs *args,
     dim3 grid((uint32_t)blocks);
     dim3 blk(256);
     switch (algo) {
-    case 1: /* SHA256: 2 chains per lane (serial rounds -> ILP; NC=4
-               spills to scratch at 217 VGPR) */
-        grid.x = (uint32_t)(((args->n_chains + 1) / 2 + 255) / 256);
-        hipLaunchKernelGGL(sha256_batch_kernel<2>, grid, blk, 0, stream,
-                           *args);
-
+    case 1: /* SHA256: serial rounds -> fill dependency-latency stalls
+               with NC interleaved chains.  NC=2 at WG 256 is the default.
+               MEC_SHA_NC=4 (64-thread launch-bounds variant, full VGPR
+               budget) measured 2.7x SLOWER (29.3 vs 10.7 ms at config #3
+               — 233 VGPR + 272 B scratch; the 4-chain schedule thrashes)
+               and is kept only as the recorded negative.  Ch/Maj/sigma
+               run as single v_bitop3 LUT ops; that changed nothing
+               either (the chain is latency-bound, not issue-bound). */
+        {
+            static const int nc = gf_env_int("MEC_SHA_NC", 2);
+            if (nc >= 4) {
+                dim3 b64(64);
+                grid.x =
+                    (uint32_t)(((args->n_chains + 3) / 4 + 63) / 64);
+                hipLaunchKernelGGL((sha256_batch_kernel<4, 64>), grid, b64,
+                                   0, stream, *args);
+            } else {
+                grid.x = (uint32_t)(((args->n_chains + 1) / 2 + 255) / 256);
+                hipLaunchKernelGGL((sha256_batch_kernel<2>), grid, blk, 0,
+                                   stream, *args);
+            }
+        }
         break;
     case 2: /* HighwayHash256 */
     case 3: /* HighwayHash256S: 2 lanes/chain (zipper pairs).  32-aligned
