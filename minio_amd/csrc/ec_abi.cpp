@@ -599,19 +599,36 @@ mec_status mec_reconstruct_batch(mec_ctx *ctx, int n, uint8_t *shards,
         return st;
     if ((st = ctx->ensure_pin(bytes)) != MEC_OK) return st;
     uint8_t *pinb = (uint8_t *)ctx->pin;
-    for (int b = 0; b < n; b++)
-        for (int s = 0; s < total; s++) {
-            uint8_t *dst = pinb + ((size_t)b * total + s) * stride;
-            memcpy(dst, shards + ((size_t)b * total + s) * shard_len,
-                   (size_t)shard_len);
-        }
-    HIP_TRY(hipMemcpyAsync(ctx->dev_a, pinb, bytes, hipMemcpyHostToDevice,
-                           ctx->stream));
-    st = reconstruct_dev_locked(ctx, n, ctx->dev_a, present, shard_len,
-                                data_only);
-    if (st != MEC_OK) return st;
-    HIP_TRY(hipMemcpyAsync(pinb, ctx->dev_a, bytes, hipMemcpyDeviceToHost,
-                           ctx->stream));
+    /* overlapped ingest (parallelReader-style): the batch is
+     * fed to the GPU reconstruct queue in chunks — chunk c+1's host pack
+     * and PCIe upload (stream2) run while chunk c's decode kernels run
+     * (stream); the event handshake keeps kernel c after upload c. */
+    const int64_t chunk = (n > 128) ? ((n + 3) / 4) : n;
+    for (int64_t b0 = 0; b0 < n; b0 += chunk) {
+        const int64_t nc = (b0 + chunk <= n) ? chunk : (n - b0);
+        uint8_t *pc = pinb + (size_t)b0 * total * stride;
+        for (int64_t b = 0; b < nc; b++)
+            for (int s = 0; s < total; s++)
+                memcpy(pc + ((size_t)b * total + s) * stride,
+                       shards + ((size_t)(b0 + b) * total + s) * shard_len,
+                       (size_t)shard_len);
+        HIP_TRY(hipMemcpyAsync((uint8_t *)ctx->dev_a +
+                                   (size_t)b0 * total * stride,
+                               pc, (size_t)nc * total * stride,
+                               hipMemcpyHostToDevice, ctx->stream2));
+        HIP_TRY(hipEventRecord(ctx->ev_h, ctx->stream2));
+        HIP_TRY(hipStreamWaitEvent(ctx->stream, ctx->ev_h, 0));
+        st = reconstruct_dev_locked(
+            ctx, (int)nc,
+            (uint8_t *)ctx->dev_a + (size_t)b0 * total * stride, present,
+            shard_len, data_only);
+        if (st != MEC_OK) return st;
+        HIP_TRY(hipMemcpyAsync(pc,
+                               (uint8_t *)ctx->dev_a +
+                                   (size_t)b0 * total * stride,
+                               (size_t)nc * total * stride,
+                               hipMemcpyDeviceToHost, ctx->stream));
+    }
     HIP_TRY(hipStreamSynchronize(ctx->stream));
     for (int b = 0; b < n; b++)
         for (int s = 0; s < total; s++) {
@@ -667,20 +684,32 @@ mec_status mec_bitrot_sum_batch(mec_ctx *ctx, int algo, int n,
         MEC_OK)
         return st;
     uint8_t *pinb = (uint8_t *)ctx->pin;
-    for (int i = 0; i < n; i++)
-        memcpy(pinb + (size_t)i * dstride, msgs + (size_t)i * msg_stride,
-               (size_t)msg_len);
-    HIP_TRY(hipMemcpyAsync(ctx->dev_a, pinb, in_bytes, hipMemcpyHostToDevice,
-                           ctx->stream));
-    HashArgs h{};
-    h.data = (const uint8_t *)ctx->dev_a;
-    h.parity = nullptr;
-    h.sums = (uint8_t *)ctx->dev_c;
-    h.row_stride = dstride;
-    h.msg_len = msg_len;
-    h.n_chains = n;
-    memcpy(h.key, kMagicHHKey, 32);
-    HIP_TRY(mec_launch_hash(algo, &h, ctx->stream));
+    /* overlapped ingest: chunk c+1's host pack + PCIe
+     * upload (stream2) overlap chunk c's hash kernel (stream) — the
+     * GPU-side analogue of parallelReader's overlapped shard reads
+     * feeding verify-on-read (cmd/erasure-decode.go:127-235). */
+    const int64_t chunk = (n > 512) ? ((n + 3) / 4) : n;
+    for (int64_t i0 = 0; i0 < n; i0 += chunk) {
+        const int64_t nc = (i0 + chunk <= n) ? chunk : (n - i0);
+        uint8_t *pc = pinb + (size_t)i0 * dstride;
+        for (int64_t i = 0; i < nc; i++)
+            memcpy(pc + (size_t)i * dstride,
+                   msgs + (size_t)(i0 + i) * msg_stride, (size_t)msg_len);
+        HIP_TRY(hipMemcpyAsync((uint8_t *)ctx->dev_a + (size_t)i0 * dstride,
+                               pc, (size_t)nc * dstride,
+                               hipMemcpyHostToDevice, ctx->stream2));
+        HIP_TRY(hipEventRecord(ctx->ev_h, ctx->stream2));
+        HIP_TRY(hipStreamWaitEvent(ctx->stream, ctx->ev_h, 0));
+        HashArgs h{};
+        h.data = (const uint8_t *)ctx->dev_a + (size_t)i0 * dstride;
+        h.parity = nullptr;
+        h.sums = (uint8_t *)ctx->dev_c + (size_t)i0 * hsz;
+        h.row_stride = dstride;
+        h.msg_len = msg_len;
+        h.n_chains = nc;
+        memcpy(h.key, kMagicHHKey, 32);
+        HIP_TRY(mec_launch_hash(algo, &h, ctx->stream));
+    }
     HIP_TRY(hipMemcpyAsync(pinb, ctx->dev_c, out_bytes, hipMemcpyDeviceToHost,
                            ctx->stream));
     HIP_TRY(hipStreamSynchronize(ctx->stream));
