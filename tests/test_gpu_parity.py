@@ -408,3 +408,53 @@ def test_fullsize_roundtrip_batch():
                     damaged[d - 1 - i] = None  # erase p rows spanning data
                 rec = e.decode_data_and_parity_blocks(damaged)
                 assert rec == shards[b]
+
+
+def test_chunked_overlap_paths_large_batches():
+    """The host-pointer reconstruct (n>128) and bitrot-sum (n>512) paths
+    chunk the batch and pipeline host pack + PCIe upload under the
+    previous chunk's kernels.  Pin bit-exactness across the
+    chunk boundaries, including a batch size that is not a multiple of
+    the chunk count."""
+    import ctypes
+    d, p, total = 4, 2, 6
+    S = 1024
+    n = 203  # > 128 -> chunked; 203 % 4 != 0 -> ragged last chunk
+    ors = oracle.RS(d, p)
+    blocks = []
+    packed = bytearray()
+    for b in range(n):
+        data = rnd(d * S, SEED + b)
+        ds = [data[i * S:(i + 1) * S] for i in range(d)]
+        shards = ds + ors.encode_blocks(ds)
+        blocks.append(shards)
+        for s in range(total):
+            packed += shards[s]
+    present = bytes([0, 1, 1, 1, 0, 1])  # rows 0 and 4 erased
+    buf = ctypes.create_string_buffer(bytes(packed), len(packed))
+    for b in range(n):
+        for s in range(total):
+            if not present[s]:
+                off = (b * total + s) * S
+                ctypes.memset(ctypes.byref(buf, off), 0, S)
+    with minio_amd.Erasure(d, p, d * S) as e:
+        minio_amd._check(minio_amd._lib.mec_reconstruct_batch(
+            e._ck(), n, buf, present, S, 0))
+        out = buf.raw
+        for b in range(n):
+            for s in range(total):
+                off = (b * total + s) * S
+                assert out[off:off + S] == blocks[b][s], (b, s)
+
+        # chunked bitrot-sum: 600 ragged messages in a strided layout
+        nmsg, mlen, mstride = 600, 549, 576
+        msgs = bytearray(nmsg * mstride)
+        for i in range(nmsg):
+            msgs[i * mstride:i * mstride + mlen] = rnd(mlen, SEED ^ i)
+        sums = e.bitrot_sum_batch(minio_amd.HIGHWAYHASH256S, bytes(msgs),
+                                  mlen, mstride, nmsg)
+        for i in range(nmsg):
+            want = oracle.bitrot_sum(
+                oracle.HIGHWAYHASH256S,
+                bytes(msgs[i * mstride:i * mstride + mlen]))
+            assert sums[i] == want, i
