@@ -458,3 +458,26 @@ def test_chunked_overlap_paths_large_batches():
                 oracle.HIGHWAYHASH256S,
                 bytes(msgs[i * mstride:i * mstride + mlen]))
             assert sums[i] == want, i
+
+
+def test_stream_boundary_cases():
+    """Deterministic edges from the reference's decode table
+    (cmd/erasure-decode_test.go:35-85): tiny objects (< one shard, < d
+    bytes), exact block-boundary offsets, single-byte reads at both ends,
+    and zero-length reads.  Streams are also pinned against the oracle's
+    writer byte-for-byte."""
+    d, p, bs = 4, 2, 8192
+    with minio_amd.Erasure(d, p, bs) as e:
+        for total_len in (1, 3, d - 1, d, 37, bs - 1, bs, bs + 1,
+                          2 * bs, 2 * bs + 17):
+            data = rnd(total_len, SEED ^ total_len)
+            streams, _ = e.encode_stream(data)
+            ostreams, _ = oracle.encode_stream(d, p, bs, data,
+                                               oracle.HIGHWAYHASH256S)
+            assert list(streams) == list(ostreams), total_len
+            cases = {(0, total_len), (0, 1), (total_len - 1, 1), (0, 0)}
+            if total_len > bs:
+                cases |= {(bs, total_len - bs), (bs - 1, 2), (bs, 1)}
+            for off, ln in sorted(cases):
+                got = e.decode_stream(list(streams), total_len, off, ln)
+                assert got == data[off:off + ln], (total_len, off, ln)
