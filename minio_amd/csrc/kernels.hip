@@ -13,10 +13,12 @@
  *    hashing is sequential per shard, parallelism comes from shards x
  *    blocks (SURVEY.md §7 hard part (b)).
  *
- * Round-1 structure: encode and hash run as back-to-back kernels on one
- * stream (data re-read by the hash kernel rides L2 for tile-sized batches).
- * The single-pass fused kernel is the planned round-2 optimization; the
- * C-ABI signature already treats the pair as one fused operation.
+ * Shipped structure: encode and hash run as two kernels, cross-batch
+ * pipelined (batch t's hash overlaps batch t+1's GF on a second stream) —
+ * measured at the joint memory wall (~4.4 TB/s).  Single-
+ * pass fused variants (fused.hip lockstep, fused2.hip producer/consumer
+ * LDS ring) are in-tree, bit-exact, and measured slower; the C-ABI
+ * signature treats the pair as one fused operation either way.
  */
 #include <hip/hip_runtime.h>
 #include <cstdint>
