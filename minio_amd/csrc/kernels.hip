@@ -1142,10 +1142,11 @@ hipError_t mec_launch_hash(int algo, const HashArgs *args,
     switch (algo) {
     case 1: /* SHA256: serial rounds -> fill dependency-latency stalls
                with NC interleaved chains.  NC=2 at WG 256 is the default.
-               MEC_SHA_NC=4 (64-thread launch-bounds variant, full VGPR
-               budget) measured 2.7x SLOWER (29.3 vs 10.7 ms at config #3
-               — 233 VGPR + 272 B scratch; the 4-chain schedule thrashes)
-               and is kept only as the recorded negative.  Ch/Maj/sigma
+               MEC_SHA_NC=4 (64-thread launch-bounds variant) measured
+               2.7x SLOWER (29.3 vs 10.7 ms at config #3) and NC=3 at
+               LB 128 measured 2.2x slower (23.2 ms) — both spill the
+               message schedule to scratch (272/208 B) and thrash; they
+               are kept only as recorded negatives.  Ch/Maj/sigma
                run as single v_bitop3 LUT ops; that changed nothing
                either (the chain is latency-bound, not issue-bound). */
         {
@@ -1156,6 +1157,12 @@ hipError_t mec_launch_hash(int algo, const HashArgs *args,
                     (uint32_t)(((args->n_chains + 3) / 4 + 63) / 64);
                 hipLaunchKernelGGL((sha256_batch_kernel<4, 64>), grid, b64,
                                    0, stream, *args);
+            } else if (nc == 3) {
+                dim3 b128(128);
+                grid.x =
+                    (uint32_t)(((args->n_chains + 2) / 3 + 127) / 128);
+                hipLaunchKernelGGL((sha256_batch_kernel<3, 128>), grid,
+                                   b128, 0, stream, *args);
             } else {
                 grid.x = (uint32_t)(((args->n_chains + 1) / 2 + 255) / 256);
                 hipLaunchKernelGGL((sha256_batch_kernel<2>), grid, blk, 0,
