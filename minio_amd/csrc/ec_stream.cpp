@@ -341,6 +341,10 @@ mec_status mec_heal_stream(mec_ctx *ctx_, const uint8_t *const *drive_bufs,
     int d = mec_ctx_d(ctx_), p = mec_ctx_p(ctx_);
     int64_t block_size = mec_ctx_block_size(ctx_);
     if (!hash_size(algo)) return MEC_ERR_INVALID_ARG;
+    if (total_length < 0) return MEC_ERR_INVALID_ARG;
+    if (total_length == 0) return MEC_OK; /* empty part: nothing to heal
+        (a 0-length object has no shard stream; without this guard the
+        ragged-last path below would index block -1) */
     Geo g = make_geo(d, p, block_size, total_length);
     const int hsz = hash_size(algo);
     const bool streaming = algo == MEC_BITROT_HIGHWAYHASH256S;

@@ -500,3 +500,23 @@ def test_reconstruct_exhaustive_patterns_ec84():
                            for i in range(total)]
                 rec = e.decode_data_and_parity_blocks(damaged)
                 assert rec == oshards, f"erased={erased}"
+
+
+def test_empty_object_paths():
+    """0-length objects: encode produces empty streams, decode of a
+    zero-length range succeeds, heal is a no-op (regression: the
+    ragged-last verify used to index block -1 for total_length==0)."""
+    d, p = 4, 2
+    with minio_amd.Erasure(d, p, 4096) as e:
+        streams, _ = e.encode_stream(b"")
+        assert all(s == b"" for s in streams)
+        assert e.decode_stream(list(streams), 0, 0, 0) == b""
+        dmg = list(streams)
+        dmg[0] = None
+        healed = e.heal_stream(dmg, 0)
+        assert healed[0] == b""
+        # whole-file algo on an empty object still produces digests
+        wstreams, wsums = e.encode_stream(b"", minio_amd.SHA256)
+        assert all(s == b"" for s in wstreams)
+        assert wsums is not None and len(wsums) == d + p
+        assert wsums[0] == oracle.bitrot_sum(oracle.SHA256, b"")
