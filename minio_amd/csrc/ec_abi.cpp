@@ -220,10 +220,22 @@ mec_status mec_ctx_create(int d, int p, int64_t block_size, int device,
                 }
             }
             (void)given;
-            if (e == hipSuccess)
-                e = hipExtStreamCreateWithCUMask(&ctx->stream, 8, mg);
-            if (e == hipSuccess)
-                e = hipExtStreamCreateWithCUMask(&ctx->stream2, 8, mh);
+            /* note: gfx950 silently IGNORES CU masks (measured flat in the
+             * r1 sweep) — the knob is a recorded no-op there.  If masked
+             * stream creation fails, fall back to plain streams: this is a
+             * tuning knob, never a correctness dependency. */
+            if (e == hipSuccess) {
+                hipError_t em =
+                    hipExtStreamCreateWithCUMask(&ctx->stream, 8, mg);
+                if (em == hipSuccess)
+                    em = hipExtStreamCreateWithCUMask(&ctx->stream2, 8, mh);
+                if (em != hipSuccess) {
+                    if (ctx->stream) (void)hipStreamDestroy(ctx->stream);
+                    ctx->stream = nullptr;
+                    ctx->stream2 = nullptr;
+                    cu_split = 0; /* plain-stream fallback below */
+                }
+            }
         }
     }
     if (cu_split <= 0) {
@@ -811,6 +823,10 @@ mec_status mec_encode_stream_gpu(mec_ctx *ctx, const uint8_t *src,
     const size_t par_bytes = (size_t)n_full * p * stride;
     const size_t sum_bytes = (size_t)n_full * total * 32;
     const size_t out_bytes = (size_t)total * n_full * pitch;
+    /* shard rows live after the stream-out region; rows must keep the 64-B
+     * base alignment the row kernels' uint4 accesses assume, and out_bytes
+     * is odd whenever pitch (32+S) is (ragged S, e.g. EC12+4) */
+    const size_t out_aligned = (out_bytes + 63) & ~(size_t)63;
     mec_status st;
     if ((st = ctx->ensure(&ctx->dev_a, &ctx->cap_a,
                           row_bytes > src_bytes ? row_bytes : src_bytes)) !=
@@ -821,10 +837,10 @@ mec_status mec_encode_stream_gpu(mec_ctx *ctx, const uint8_t *src,
     if ((st = ctx->ensure(&ctx->dev_c, &ctx->cap_c, sum_bytes)) != MEC_OK)
         return st;
     if ((st = ctx->ensure(&ctx->dev_d, &ctx->cap_d,
-                          out_bytes + row_bytes)) != MEC_OK)
+                          out_aligned + row_bytes)) != MEC_OK)
         return st;
     /* dev_d holds [stream out | shard rows]; dev_a stages the packed src */
-    uint8_t *dev_rows = (uint8_t *)ctx->dev_d + out_bytes;
+    uint8_t *dev_rows = (uint8_t *)ctx->dev_d + out_aligned;
     HIP_TRY(hipMemcpyAsync(ctx->dev_a, src, src_bytes, hipMemcpyHostToDevice,
                            ctx->stream));
     ScatterArgs sa{};

@@ -83,3 +83,58 @@ def test_c_example_compiles_and_links():
         r = subprocess.run([exe], capture_output=True, text=True)
         assert r.returncode != 0  # no GPU here -> loud failure
         assert "MI355X required" in r.stderr or "no HIP device" in r.stderr
+
+
+def test_distribution_permutation_roundtrip():
+    """Pins the reference's shard-distribution permutation (the seam a
+    deeper shim must honor; INTEGRATION.md "Shard distribution").
+
+    hashOrder (cmd/erasure-metadata-utils.go:178): Distribution[k] is the
+    1-based LOGICAL shard index physical drive k holds — a rotation of
+    1..n started at crc32(key)%n.  Golden vectors are the reference's own
+    TestHashOrder table (cmd/erasure-metadata-utils_test.go:115-130).
+    """
+    import zlib
+
+    def hash_order(key: str, n: int):
+        if n <= 0:
+            return None
+        crc = zlib.crc32(key.encode("utf-8")) & 0xFFFFFFFF
+        start = crc % n
+        return [1 + ((start + i) % n) for i in range(1, n + 1)]
+
+    golden = {
+        "object": [14, 15, 16, 1, 2, 3, 4, 5, 6, 7, 8, 9, 10, 11, 12, 13],
+        "The Shining Script <v1>.pdf":
+            [16, 1, 2, 3, 4, 5, 6, 7, 8, 9, 10, 11, 12, 13, 14, 15],
+        "SHØRT": [11, 12, 13, 14, 15, 16, 1, 2, 3, 4, 5, 6, 7, 8, 9, 10],
+        "a/b/c/": [3, 4, 5, 6, 7, 8, 9, 10, 11, 12, 13, 14, 15, 16, 1, 2],
+        "/a/b/c": [6, 7, 8, 9, 10, 11, 12, 13, 14, 15, 16, 1, 2, 3, 4, 5],
+    }
+    for key, want in golden.items():
+        assert hash_order(key, 16) == want, key
+    assert hash_order("x", -1) is None and hash_order("x", 0) is None
+
+    # logical -> physical -> logical round trip (identity and rotated):
+    # writer: physical drive k receives logical shard distribution[k]-1;
+    # reader: logical[distribution[k]-1] = physical[k]
+    n = 12
+    logical = [f"shard{j}".encode() for j in range(n)]
+    for key in ["object", "a/b/c/", "bucket/deep/key"]:
+        dist = hash_order(key, n)
+        physical = [logical[dist[k] - 1] for k in range(n)]
+        back = [None] * n
+        for k in range(n):
+            back[dist[k] - 1] = physical[k]
+        assert back == logical, key
+    # identity case: a key whose rotation lands at 0
+    for key in ("object", "obj2", "obj17", "a", "b", "c", "dd", "zz9"):
+        dist = hash_order(key, n)
+        if dist == list(range(1, n + 1)):
+            break
+    # (identity occurs for 1/n of keys; mapping formula covers it anyway)
+    physical = [logical[d - 1] for d in dist]
+    back = [None] * n
+    for k in range(n):
+        back[dist[k] - 1] = physical[k]
+    assert back == logical

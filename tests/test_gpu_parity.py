@@ -482,6 +482,23 @@ def test_stream_boundary_cases():
                 got = e.decode_stream(list(streams), total_len, off, ln)
                 assert got == data[off:off + ln], (total_len, off, ln)
 
+    # ragged odd-pitch geometries with FULL blocks: odd S makes the
+    # per-entry pitch (32+S) odd, so the stream-out region's byte count is
+    # not 16-B aligned — the device row buffer placed after it must round
+    # up to keep the row kernels' uint4 alignment invariant (regression:
+    # scatter/interleave used misaligned uint4 at d=4,p=2,S=10)
+    for (d2, p2, bs2) in ((4, 2, 40), (3, 2, 51)):
+        with minio_amd.Erasure(d2, p2, bs2) as e2:
+            for nfull in (1, 2, 5):
+                total_len = nfull * bs2 + 7
+                data = rnd(total_len, SEED ^ (d2 * 131 + total_len))
+                streams, _ = e2.encode_stream(data)
+                ostreams, _ = oracle.encode_stream(d2, p2, bs2, data,
+                                                   oracle.HIGHWAYHASH256S)
+                assert list(streams) == list(ostreams), (d2, p2, nfull)
+                got = e2.decode_stream(list(streams), total_len, 0, total_len)
+                assert got == data, (d2, p2, nfull)
+
 
 def test_reconstruct_exhaustive_patterns_ec84():
     """Every erasure pattern the EC8+4 geometry can see: all C(12,k)
@@ -520,3 +537,52 @@ def test_empty_object_paths():
         assert all(s == b"" for s in wstreams)
         assert wsums is not None and len(wsums) == d + p
         assert wsums[0] == oracle.bitrot_sum(oracle.SHA256, b"")
+
+
+# ---- off-default experiment knobs (env-latched -> subprocess) -----------
+
+_KNOB_CHECK = r"""
+import os, sys
+sys.path.insert(0, os.environ["MEC_TEST_REPO"])
+import minio_amd, oracle
+SEED = 0x6D696E696F
+d, p, bs = 8, 4, 64 * 8 * 1024   # S = 64 KiB: %1024==0 (fused2), >=512 (LDS)
+with minio_amd.Erasure(d, p, bs) as e:
+    for n, seed in ((7, 1), (32, 2)):
+        blocks = [oracle.fill_random(bs, SEED + seed * 100 + b)
+                  for b in range(n)]
+        shards, sums = e.encode_batch(b"".join(blocks), bs, n,
+                                      minio_amd.HIGHWAYHASH256S)
+        ors = oracle.RS(d, p)
+        for b in range(n):
+            osh = ors.encode_data(blocks[b])
+            assert shards[b] == osh, b
+            for s in range(d + p):
+                want = oracle.bitrot_sum(oracle.HIGHWAYHASH256S, osh[s])
+                assert sums[b][s] == want, (b, s)
+    # ragged message lengths through the knob'd hash kernel too
+    msgs = b"".join(oracle.fill_random(577, SEED + i) for i in range(65))
+    sums = e.bitrot_sum_batch(minio_amd.HIGHWAYHASH256S, msgs, 577, 577, 65)
+    for i in range(65):
+        assert sums[i] == oracle.bitrot_sum(
+            oracle.HIGHWAYHASH256S, msgs[i * 577:(i + 1) * 577]), i
+print("KNOB_OK")
+"""
+
+
+@pytest.mark.parametrize("knob", ["MEC_HH_LDS", "MEC_FUSED2", "MEC_FUSED",
+                                  "MEC_HH_WG"])
+def test_knob_variants_bit_exact(knob):
+    """The in-tree experiment knobs are env-latched at first
+    use, so each variant runs in a subprocess.  Every knob'd kernel must
+    stay bit-exact vs the oracle (covers the MEC_HH_LDS hand-placed
+    s_waitcnt double-buffer handoff, which no default-path test reaches)."""
+    import subprocess
+    env = dict(os.environ)
+    env[knob] = "512" if knob == "MEC_HH_WG" else "1"
+    env["MEC_TEST_REPO"] = os.path.dirname(HERE)
+    r = subprocess.run([os.environ.get("PYTHON", "python3"), "-c",
+                        _KNOB_CHECK], env=env, capture_output=True,
+                       text=True, timeout=600)
+    assert r.returncode == 0 and "KNOB_OK" in r.stdout, (
+        knob, r.returncode, r.stdout[-2000:], r.stderr[-2000:])

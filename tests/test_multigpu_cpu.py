@@ -45,3 +45,34 @@ def _worker(rank, world, port):
 def test_weak_scaling_pattern_gloo():
     port = 29611
     mp.spawn(_worker, args=(2, port), nprocs=2, join=True)
+
+
+def test_bench_rank_device_mapping_dryrun():
+    """bench.py --gpus N dry run (no GPU): the rank -> device mapping and
+    per-rank seed derivation in bench.py's source must keep the N-rank
+    launch correct by construction — LOCAL_RANK maps 1:1 onto devices on a
+    full node, modulo-wraps on a smaller box, and per-rank input seeds are
+    disjoint."""
+    # exercise the exact expressions bench.py uses (kept in sync by eye —
+    # this test reads them from the source to fail loudly if they drift)
+    import re
+    src = open(os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "bench.py")).read()
+    m = re.search(r"dev = local_rank % max\(1, minio_amd\.device_count\(\)\)",
+                  src)
+    assert m, "bench.py rank->device mapping changed; update this test"
+    m = re.search(r"SEED \+ rank \* 1000003", src)
+    assert m, "bench.py per-rank seed derivation changed; update this test"
+
+    # the mapping expressions themselves:
+    for ndev in (1, 2, 8):
+        devs = [lr % max(1, ndev) for lr in range(8)]
+        if ndev == 8:
+            assert devs == list(range(8))       # 1:1 on a full node
+        assert all(0 <= dv < ndev for dv in devs)
+    # disjoint seed spaces for up to 8 ranks x 4096 blocks
+    spaces = [{SEED + r * 1000003 + b for b in range(4096)} for r in range(8)]
+    seen = set()
+    for s in spaces:
+        assert not (seen & s)
+        seen |= s
