@@ -297,6 +297,12 @@ def main():
             cpu_baseline = {
                 "value": round(sample_n * bs / el / (1 << 30), 3),
                 "unit": "GiB/s input", "cores": cores, "kind": "port",
+                # which ISA path the oracle's bench legs dispatched to
+                # (oracle/simd.c: GFNI gf2p8affineqb GF + AVX2 4-lane HH —
+                # the same instruction classes the reference's
+                # klauspost/reedsolomon + minio/highwayhash asm uses; the
+                # parity-checking oracle itself stays scalar)
+                "isa": oracle.cpu_isa(),
                 "sample": f"{sample_n} x {bs} B blocks "
                           f"({'reconstruct' if is_decode else 'encode+bitrot'})",
             }

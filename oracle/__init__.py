@@ -42,6 +42,16 @@ _lib.mo_cpu_encode_bench.argtypes = [
     ctypes.c_int, ctypes.c_int, ctypes.c_size_t, ctypes.c_int,
     ctypes.c_int, ctypes.c_int, ctypes.c_uint64,
 ]
+_lib.mo_cpu_isa.restype = ctypes.c_char_p
+_lib.mo_cpu_isa.argtypes = []
+_lib.mo_gal_mul_xor_fast.argtypes = [
+    ctypes.c_uint8, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_size_t,
+]
+_lib.mo_hh256_fast.argtypes = [
+    ctypes.c_char_p, ctypes.c_char_p, ctypes.c_size_t, ctypes.c_char_p,
+]
+_lib.mo_gf_mul.restype = ctypes.c_uint8
+_lib.mo_gf_mul.argtypes = [ctypes.c_uint8, ctypes.c_uint8]
 _lib.mo_cpu_reconstruct_bench.restype = ctypes.c_double
 _lib.mo_cpu_reconstruct_bench.argtypes = [
     ctypes.c_int, ctypes.c_int, ctypes.c_size_t, ctypes.c_int,
@@ -184,3 +194,22 @@ def encode_stream(d: int, p: int, block_size: int, data: bytes, algo: int):
     if algo == HIGHWAYHASH256S:
         return streams, None
     return streams, [bitrot_sum(algo, b"".join(w)) for w in whole]
+
+
+def cpu_isa() -> str:
+    """ISA path the SIMD bench legs dispatch to (simd.c): "gfni+avx2",
+    "avx2", or "scalar"."""
+    return _lib.mo_cpu_isa().decode()
+
+
+def gal_mul_xor_fast(c: int, data: bytes, acc: bytes) -> bytes:
+    """SIMD bench-leg GF op: returns acc ^= c*data (bytes)."""
+    out = ctypes.create_string_buffer(acc, len(acc))
+    _lib.mo_gal_mul_xor_fast(c, data, out, len(data))
+    return out.raw[:len(data)]
+
+
+def hh256_fast(key32: bytes, msg: bytes) -> bytes:
+    out = ctypes.create_string_buffer(32)
+    _lib.mo_hh256_fast(key32, msg, len(msg), out)
+    return out.raw

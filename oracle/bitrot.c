@@ -48,6 +48,20 @@ void mo_bitrot_sum(int algo, const uint8_t *msg, size_t len, uint8_t *out) {
     }
 }
 
+/* bench-leg dispatch: SIMD hash where it exists (HighwayHash), scalar
+ * otherwise (SHA-256/BLAKE2b are not on the bench legs' hot configs) */
+void mo_bitrot_sum_fast(int algo, const uint8_t *msg, size_t len,
+                        uint8_t *out) {
+    switch (algo) {
+    case MO_BITROT_HIGHWAYHASH256:
+    case MO_BITROT_HIGHWAYHASH256S:
+        mo_hh256_fast(MAGIC_HH_KEY, msg, len, out);
+        break;
+    default:
+        mo_bitrot_sum(algo, msg, len, out);
+    }
+}
+
 /* xoshiro256** — seeded synthetic inputs (seed stated in bench output) */
 typedef struct { uint64_t s[4]; } xo_state;
 
@@ -134,10 +148,10 @@ double mo_cpu_encode_bench(int d, int p, size_t block_len, int n_blocks,
             shards[k] = data + ((size_t)b * d + k) * shard_len;
         for (int i = 0; i < p; i++)
             shards[d + i] = parity + ((size_t)b * p + i) * shard_len;
-        mo_rs_encode(&rs, shards, shard_len);
+        mo_rs_encode_fast(&rs, shards, shard_len);
         for (int s = 0; s < total; s++)
-            mo_bitrot_sum(algo, shards[s], shard_len,
-                          sums + ((size_t)b * total + s) * hsz);
+            mo_bitrot_sum_fast(algo, shards[s], shard_len,
+                               sums + ((size_t)b * total + s) * hsz);
     }
     double el = now_sec() - t0;
     /* keep the compiler honest */

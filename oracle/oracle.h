@@ -95,6 +95,19 @@ void mo_fill_random(uint8_t *buf, size_t n, uint64_t seed);
  * Fused encode+bitrot of n_blocks blocks of block_len random bytes
  * (seeded xoshiro256**), EC d+p, per-shard digest with `algo`.
  * Runs on `threads` OpenMP threads.  Returns elapsed seconds. */
+/* ---- SIMD bench legs (simd.c): runtime-dispatched GFNI/AVX2/scalar.
+ * BENCH ONLY — the parity checker stays the scalar restatement; these are
+ * pinned bit-exact against it by test_simd_matches_scalar. */
+const char *mo_cpu_isa(void);
+void mo_gal_mul_xor_fast(uint8_t c, const uint8_t *in, uint8_t *out,
+                         size_t n);
+void mo_rs_encode_fast(const mo_rs *rs, uint8_t *const *shards,
+                       size_t shard_len);
+void mo_hh256_fast(const uint8_t key[32], const uint8_t *msg, size_t len,
+                   uint8_t out[32]);
+void mo_bitrot_sum_fast(int algo, const uint8_t *msg, size_t len,
+                        uint8_t *out);
+
 double mo_cpu_encode_bench(int d, int p, size_t block_len, int n_blocks,
                            int algo, int threads, uint64_t seed);
 /* Same for reconstruction with the first n_erased shards erased. */

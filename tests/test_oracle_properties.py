@@ -99,3 +99,31 @@ def test_encode_stream_layout():
             assert h == oracle.bitrot_sum(oracle.HIGHWAYHASH256S, sh)
             off += 32 + len(shards[i])
         assert off == len(st)
+
+
+def test_simd_matches_scalar():
+    """The SIMD bench legs (oracle/simd.c; BENCH ONLY, never the checker)
+    must be bit-identical to the scalar restatement: GF constant-multiply
+    for every coefficient, and the AVX2 HighwayHash main loop across
+    lengths covering every remainder residue."""
+    import ctypes
+    isa = oracle.cpu_isa()
+    assert isa in ("gfni+avx2", "avx2", "scalar")
+    # GF: all 256 coefficients over a buffer covering all byte values,
+    # with a ragged tail exercising the scalar cleanup
+    n = 256 * 3 + 7
+    data = bytes((i * 37 + 11) & 0xFF for i in range(n))
+    acc0 = bytes((i * 101 + 5) & 0xFF for i in range(n))
+    for c in range(256):
+        want = bytes(a ^ oracle._lib.mo_gf_mul(c, b)
+                     for a, b in zip(acc0, data)) if c else acc0
+        got = oracle.gal_mul_xor_fast(c, data, acc0)
+        assert got == want, f"coef {c} (isa={isa})"
+    # HH: magic key, lengths covering 0..63 plus larger odd sizes
+    key = bytes.fromhex(
+        "4be734fa8e238acd263e83e6bb96855204 0f935da39f441497e09d1322de36a0"
+        .replace(" ", ""))
+    for ln in list(range(0, 64)) + [100, 1024, 4096 + 31, 87382]:
+        msg = oracle.fill_random(ln, 0xABCDEF + ln)
+        assert oracle.hh256_fast(key, msg) == oracle.bitrot_sum(
+            oracle.HIGHWAYHASH256S, msg), f"len {ln} (isa={isa})"
