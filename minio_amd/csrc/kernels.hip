@@ -509,6 +509,188 @@ __global__ void __launch_bounds__(512) hh256_batch_kernel(HashArgs a) {
     }
 }
 
+/* ---- HighwayHash-256, 4 lanes per chain (round-2 default candidate) ----
+ *
+ * Why 4 GPU lanes per chain instead of the r1 pair-lane split: at the
+ * headline chain count (12288) the pair kernel is 384 waves — only 96 CUs
+ * at WG 256 — and r1's ISA shows ~69 VALU/packet per lane (64-bit adds,
+ * zipper perms, word re-packing), so each wave alone on its SIMD pays
+ * dependency latency (~265 cyc/packet measured vs 138 issue floor) while
+ * only ~96 CUs' worth of load bandwidth is engaged (the WG-512 "2
+ * waves/SIMD" variant measured WORSE because it halved the engaged CUs —
+ * per-CU global-load bandwidth ~10 B/cyc is the real wall).  Splitting
+ * each chain over 4 lanes (one HighwayHash lane each):
+ *  - doubles the wave count (768 waves -> ~192 CUs engaged),
+ *  - cuts per-lane work to ~27 VALU/packet (zipper needs only the
+ *    partner's v1/v0 HIGH word: one quad-perm DPP mov per phase, and the
+ *    even/odd zipper variants collapse to one code path with a per-lane
+ *    v_perm selector),
+ *  - loads 8 B/lane/packet (dwordx2), same coalescing (a chain's 4 lanes
+ *    cover its 32-B packet contiguously).
+ */
+__device__ __forceinline__ uint32_t dpp_swap1(uint32_t v) {
+    /* value from lane ^ 1 (zipper pair partner); quad_perm [1,0,3,2] */
+    return (uint32_t)__builtin_amdgcn_mov_dpp((int)v, 0xB1, 0xF, 0xF, true);
+}
+__device__ __forceinline__ uint32_t dpp_swap2(uint32_t v) {
+    /* value from lane ^ 2 (finalization permute); quad_perm [2,3,0,1] */
+    return (uint32_t)__builtin_amdgcn_mov_dpp((int)v, 0x4E, 0xF, 0xF, true);
+}
+
+struct HH1 {
+    uint64_t v0, v1, mul0, mul1; /* this lane's single HighwayHash lane */
+};
+
+/* One packet update for HH lane j (lane parity selects the zipper hi
+ * selector; S1/S2 are parity-independent — see zip_even/zip_odd above). */
+__device__ __forceinline__ void hh1_update(HH1 &s, uint64_t w, uint32_t S3) {
+    s.v1 += s.mul0 + w;
+    s.mul0 ^= (s.v1 & 0xffffffffull) * (s.v0 >> 32);
+    s.v0 += s.mul1;
+    s.mul1 ^= (s.v0 & 0xffffffffull) * (s.v1 >> 32);
+    {
+        uint32_t own_lo = (uint32_t)s.v1, own_hi = (uint32_t)(s.v1 >> 32);
+        uint32_t p_hi = dpp_swap1(own_hi);
+        uint32_t lo = permb(own_hi, own_lo, 0x05020C03u) |
+                      permb(0u, p_hi, 0x0C0C000Cu);
+        uint32_t hi = permb(p_hi, own_lo, S3);
+        s.v0 += ((uint64_t)hi << 32) | lo;
+    }
+    {
+        uint32_t own_lo = (uint32_t)s.v0, own_hi = (uint32_t)(s.v0 >> 32);
+        uint32_t p_hi = dpp_swap1(own_hi);
+        uint32_t lo = permb(own_hi, own_lo, 0x05020C03u) |
+                      permb(0u, p_hi, 0x0C0C000Cu);
+        uint32_t hi = permb(p_hi, own_lo, S3);
+        s.v1 += ((uint64_t)hi << 32) | lo;
+    }
+}
+
+template <bool RAGGED>
+__global__ void __launch_bounds__(256) hh256_batch4_kernel(HashArgs a) {
+    const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t chain = tid >> 2;
+    const int j = (int)(tid & 3); /* this lane's HighwayHash lane index */
+    if (chain >= a.n_chains) return;
+    int64_t sum_idx;
+    const uint8_t *mp = chain_ptr(a, chain, sum_idx) + 8 * j;
+    const uint32_t S3 = (j & 1) ? 0x07000601u : 0x00070106u;
+
+    const uint64_t init0[4] = {0xdbe6d5d5fe4cce2full, 0xa4093822299f31d0ull,
+                               0x13198a2e03707344ull, 0x243f6a8885a308d3ull};
+    const uint64_t init1[4] = {0x3bd39e10cb0ef593ull, 0xc0acf169b5f18a8cull,
+                               0xbe5466cf34e90c6cull, 0x452821e638d01377ull};
+    HH1 s;
+    s.mul0 = init0[j];
+    s.mul1 = init1[j];
+    s.v0 = init0[j] ^ a.key[j];
+    s.v1 = init1[j] ^ ((a.key[j] >> 32) | (a.key[j] << 32));
+
+    int64_t len = a.msg_len;
+    constexpr int DP = 16; /* prefetch depth (packets); 8 B each */
+    uint64_t qa[DP], qb[DP];
+#define HH4_LOAD(Q)                                                          \
+    {                                                                        \
+        _Pragma("unroll") for (int t = 0; t < DP; t++)                       \
+            Q[t] = *(const uint64_t *)(mp + 32 * t);                         \
+        mp += 32 * DP;                                                       \
+    }
+#define HH4_COMP(Q)                                                          \
+    {                                                                        \
+        _Pragma("unroll") for (int t = 0; t < DP; t++)                       \
+            hh1_update(s, Q[t], S3);                                         \
+    }
+    __builtin_amdgcn_s_setprio(1);
+    if (len >= 32 * DP) {
+        HH4_LOAD(qa)
+        len -= 32 * DP;
+        while (len >= 2 * 32 * DP) {
+            HH4_LOAD(qb)
+            HH4_COMP(qa)
+            HH4_LOAD(qa)
+            HH4_COMP(qb)
+            len -= 2 * 32 * DP;
+        }
+        if (len >= 32 * DP) {
+            HH4_LOAD(qb)
+            HH4_COMP(qa)
+            HH4_COMP(qb)
+            len -= 32 * DP;
+        } else {
+            HH4_COMP(qa)
+        }
+    }
+#undef HH4_LOAD
+#undef HH4_COMP
+    __builtin_amdgcn_s_setprio(0);
+    while (len >= 32) {
+        uint64_t w = *(const uint64_t *)mp;
+        hh1_update(s, w, S3);
+        mp += 32;
+        len -= 32;
+    }
+    if (RAGGED && len > 0) {
+        /* UpdateRemainder (same semantics as hh256_batch_kernel's tail);
+         * each lane builds the padded 32-B packet privately and consumes
+         * its own 8-B word — tail-only cost */
+        const int mod32 = (int)len;
+        const int mod4 = mod32 & 3;
+        const uint8_t *tail_msg = mp - 8 * j;
+        s.v0 += ((uint64_t)mod32 << 32) + (uint64_t)mod32;
+        {
+            uint32_t h0 = (uint32_t)s.v1;
+            uint32_t h1 = (uint32_t)(s.v1 >> 32);
+            s.v1 = (uint32_t)((h0 << mod32) | (h0 >> (32 - mod32)));
+            s.v1 |= (uint64_t)((h1 << mod32) | (h1 >> (32 - mod32))) << 32;
+        }
+        uint8_t packet[32];
+#pragma unroll
+        for (int i = 0; i < 32; i++) packet[i] = 0;
+        for (int i = 0; i < (mod32 & ~3); i++) packet[i] = tail_msg[i];
+        const uint8_t *rem = tail_msg + (mod32 & ~3);
+        if (mod32 & 16) {
+            for (int i = 0; i < 4; i++)
+                packet[28 + i] = rem[i + mod4 - 4];
+        } else if (mod4) {
+            packet[16] = rem[0];
+            packet[17] = rem[mod4 >> 1];
+            packet[18] = rem[mod4 - 1];
+        }
+        uint64_t w = 0;
+        for (int bt = 7; bt >= 0; bt--) w = (w << 8) | packet[8 * j + bt];
+        hh1_update(s, w, S3);
+    }
+    /* 10 permute-update rounds: permuted[j] = rot32(v0[j ^ 2]) — rot32 of
+     * the cross-pair partner is just its halves swapped */
+#pragma unroll 1
+    for (int r = 0; r < 10; r++) {
+        uint32_t p_lo = dpp_swap2((uint32_t)s.v0);
+        uint32_t p_hi = dpp_swap2((uint32_t)(s.v0 >> 32));
+        hh1_update(s, ((uint64_t)p_lo << 32) | p_hi, S3);
+    }
+    /* modular reduction: lanes {0,1} emit hash[0..1], {2,3} hash[2..3].
+     * m for the pair needs a2 = s0 = partner-even's (v1+mul1):
+     *   even lane j: out = t_j ^ (s_j << 1) ^ (s_j << 2)
+     *   odd lane j:  out = t_j ^ ((s_j' << 1) | (s_e >> 63))
+     *                          ^ ((s_j' << 2) | (s_e >> 62)),
+     *   s_j' = s_j & 0x3fff..., s_e = even partner's s. */
+    {
+        uint64_t t = s.v0 + s.mul0;
+        uint64_t sv = s.v1 + s.mul1;
+        uint32_t se_lo = dpp_swap1((uint32_t)sv);
+        uint32_t se_hi = dpp_swap1((uint32_t)(sv >> 32));
+        uint64_t se = ((uint64_t)se_hi << 32) | se_lo; /* partner's s */
+        uint64_t out;
+        if ((j & 1) == 0) {
+            out = t ^ (sv << 1) ^ (sv << 2);
+        } else {
+            uint64_t a3 = sv & 0x3fffffffffffffffull;
+            out = t ^ ((a3 << 1) | (se >> 63)) ^ ((a3 << 2) | (se >> 62));
+        }
+        *(uint64_t *)(a.sums + sum_idx * 32 + 8 * j) = out;
+    }
+}
+
 /* ---- LDS-staged HighwayHash (MEC_HH_LDS=1) -----------------------------
  *
  * Same chain math as hh256_batch_kernel (pair-lanes, v_perm zipper), but
@@ -1151,6 +1333,27 @@ hipError_t mec_launch_hash(int algo, const HashArgs *args,
                either (the chain is latency-bound, not issue-bound). */
         {
             static const int nc = gf_env_int("MEC_SHA_NC", 2);
+            /* MEC_SHA_WG=512: 8-wave workgroups -> 2 waves/SIMD on half
+             * the CUs.  SHA's round chain is dependency-latency-bound
+             * (~17 cyc/instr at 1 wave/SIMD, r1) and its bandwidth need
+             * is low (~0.5 TB/s at config #3), so trading engaged CUs
+             * for co-resident waves that fill each other's stalls is the
+             * remaining occupancy lever (r2 probe). */
+            static const int swg = gf_env_int("MEC_SHA_WG", 256);
+            if (swg >= 512 && nc == 2) {
+                dim3 b512(512);
+                grid.x = (uint32_t)(((args->n_chains + 1) / 2 + 511) / 512);
+                hipLaunchKernelGGL((sha256_batch_kernel<2, 512>), grid,
+                                   b512, 0, stream, *args);
+                break;
+            }
+            if (swg >= 512 && nc == 1) {
+                dim3 b512(512);
+                grid.x = (uint32_t)((args->n_chains + 511) / 512);
+                hipLaunchKernelGGL((sha256_batch_kernel<1, 512>), grid,
+                                   b512, 0, stream, *args);
+                break;
+            }
             if (nc >= 4) {
                 dim3 b64(64);
                 grid.x =
@@ -1185,6 +1388,20 @@ hipError_t mec_launch_hash(int algo, const HashArgs *args,
         {
             static const int use_lds = gf_env_int("MEC_HH_LDS", 0);
             static const int wg = gf_env_int("MEC_HH_WG", 256);
+            static const int hh4 = gf_env_int("MEC_HH4", 1);
+            if (hh4 && !use_lds) {
+                /* 4-lane-per-chain kernel (see hh256_batch4_kernel):
+                 * 2x the waves of the pair kernel -> ~2x engaged CUs */
+                dim3 hblk(256);
+                grid.x = (uint32_t)((args->n_chains * 4 + 255) / 256);
+                if (args->msg_len % 32 == 0)
+                    hipLaunchKernelGGL((hh256_batch4_kernel<false>), grid,
+                                       hblk, 0, stream, *args);
+                else
+                    hipLaunchKernelGGL((hh256_batch4_kernel<true>), grid,
+                                       hblk, 0, stream, *args);
+                break;
+            }
             if (use_lds && args->msg_len >= 512) {
                 dim3 hblk(256);
                 grid.x = (uint32_t)((args->n_chains + 127) / 128);

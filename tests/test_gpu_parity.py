@@ -579,12 +579,18 @@ with minio_amd.Erasure(d, p, bs) as e:
     for i in range(65):
         assert sums[i] == oracle.bitrot_sum(
             oracle.HIGHWAYHASH256S, msgs[i * 577:(i + 1) * 577]), i
+    # SHA leg (covers the MEC_SHA_* launch variants)
+    smsgs = b"".join(oracle.fill_random(2048, SEED * 2 + i) for i in range(40))
+    ssums = e.bitrot_sum_batch(minio_amd.SHA256, smsgs, 2048, 2048, 40)
+    for i in range(40):
+        assert ssums[i] == oracle.bitrot_sum(
+            oracle.SHA256, smsgs[i * 2048:(i + 1) * 2048]), i
 print("KNOB_OK")
 """
 
 
 @pytest.mark.parametrize("knob", ["MEC_HH_LDS", "MEC_FUSED2", "MEC_FUSED",
-                                  "MEC_HH_WG"])
+                                  "MEC_HH_WG", "MEC_HH4_OFF", "MEC_SHA_WG"])
 def test_knob_variants_bit_exact(knob):
     """The in-tree experiment knobs are env-latched at first
     use, so each variant runs in a subprocess.  Every knob'd kernel must
@@ -592,7 +598,12 @@ def test_knob_variants_bit_exact(knob):
     s_waitcnt double-buffer handoff, which no default-path test reaches)."""
     import subprocess
     env = dict(os.environ)
-    env[knob] = "512" if knob == "MEC_HH_WG" else "1"
+    if knob in ("MEC_HH_WG", "MEC_SHA_WG"):
+        env[knob] = "512"
+    elif knob == "MEC_HH4_OFF":
+        env["MEC_HH4"] = "0"   # the r1 pair-lane kernel
+    else:
+        env[knob] = "1"
     env["MEC_TEST_REPO"] = os.path.dirname(HERE)
     r = subprocess.run([os.environ.get("PYTHON", "python3"), "-c",
                         _KNOB_CHECK], env=env, capture_output=True,
