@@ -308,7 +308,13 @@ static mec_status encode_dev_locked(mec_ctx *ctx, int n, const void *data_dev,
         fa.shard_len = S_call;
         fa.n = n;
         memcpy(fa.key, kMagicHHKey, 32);
-        hipError_t he = mec_launch_fused2_encode_hh(d, p, &fa, ctx->stream);
+        hipError_t he = mec_launch_fused3_encode_hh(d, p, &fa, ctx->stream);
+        if (he == hipSuccess) return MEC_OK;
+        if (he != hipErrorNotSupported) {
+            set_err("fused3_encode_hh", he);
+            return MEC_ERR_HIP;
+        }
+        he = mec_launch_fused2_encode_hh(d, p, &fa, ctx->stream);
         if (he == hipSuccess) return MEC_OK;
         if (he != hipErrorNotSupported) {
             set_err("fused2_encode_hh", he);
@@ -402,6 +408,30 @@ mec_status mec_encode_batch_dev_pipe(mec_ctx *ctx, int n,
     /* this buffer slot's previous hash must be drained before GF rewrites
      * the parity buffer */
     HIP_TRY(hipStreamWaitEvent(ctx->stream, ctx->ev_pipe[idx], 0));
+    /* single-pass fused kernel (r2 default): one launch does GF + hash at
+     * 1.5 B/input byte — cross-batch pipelining degenerates to back-to-
+     * back fused launches, which is what we want (the fused kernel is
+     * memory-bound; overlapping two of them cannot beat sequential) */
+    if (algo == MEC_BITROT_HIGHWAYHASH256 ||
+        algo == MEC_BITROT_HIGHWAYHASH256S) {
+        FusedArgs fa{};
+        fa.data = (const uint8_t *)data_dev;
+        fa.parity = (uint8_t *)parity_dev;
+        fa.sums = (uint8_t *)sums_dev;
+        fa.row_stride = ctx->stride;
+        fa.shard_len = S_call;
+        fa.n = n;
+        memcpy(fa.key, kMagicHHKey, 32);
+        hipError_t he = mec_launch_fused3_encode_hh(d, p, &fa, ctx->stream);
+        if (he == hipSuccess) {
+            HIP_TRY(hipEventRecord(ctx->ev_pipe[idx], ctx->stream));
+            return MEC_OK;
+        }
+        if (he != hipErrorNotSupported) {
+            set_err("fused3_encode_hh", he);
+            return MEC_ERR_HIP;
+        }
+    }
     {
         GfEncArgs ea{};
         ea.data = (const uint8_t *)data_dev;

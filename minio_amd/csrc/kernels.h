@@ -91,6 +91,8 @@ hipError_t mec_launch_stream_interleave(const InterleaveArgs *args,
                                         hipStream_t stream);
 hipError_t mec_launch_fused_encode_hh(int d, int p, const FusedArgs *args,
                                       hipStream_t stream);
+hipError_t mec_launch_fused3_encode_hh(int d, int p, const FusedArgs *args,
+                                       hipStream_t stream);
 hipError_t mec_launch_fused2_encode_hh(int d, int p, const FusedArgs *args,
                                        hipStream_t stream);
 hipError_t mec_launch_gf_matmul(const GfMatmulArgs *args, int n_dst, int n,
