@@ -119,16 +119,19 @@ __device__ __forceinline__ int lds_poll(int *flag, int want) {
 
 } // namespace fused3
 
-template <int D, int P, const uint8_t (&MAT)[P][D]>
-__global__ void __launch_bounds__(512) fused3_encode_hh_kernel(FusedArgs a) {
+template <int D, int P, const uint8_t (&MAT)[P][D], int WAVES = 8,
+          int NPROD_T = 4>
+__global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
+    FusedArgs a) {
     using namespace fused3;
     constexpr int TOT = D + P;
-    constexpr int G = 64 / TOT;        /* blocks per workgroup */
+    /* blocks per workgroup: consumer lanes / (4 lanes per chain x TOT) */
+    constexpr int G = (WAVES - NPROD_T) * 16 / TOT;
     constexpr int TILE = 1024;         /* bytes per shard per ring tile */
     constexpr int ROW = TILE + 16;     /* bank-skewed LDS row */
     constexpr int RING = 2;
     constexpr int SLOT = G * TOT * ROW;
-    constexpr int NPROD = 4;           /* producer waves (first half) */
+    constexpr int NPROD = NPROD_T;     /* producer waves (first half) */
     __shared__ uint8_t lds[RING * SLOT + 64];
     int *flags = (int *)&lds[RING * SLOT];
     /* flags[0..1]=ready epoch, [2..3]=cons epoch, [4..5]=prod arrivals,
@@ -275,7 +278,7 @@ __global__ void __launch_bounds__(512) fused3_encode_hh_kernel(FusedArgs a) {
             int prev = __hip_atomic_fetch_add(&flags[6 + slot], 1,
                                               __ATOMIC_RELAXED,
                                               __HIP_MEMORY_SCOPE_WORKGROUP);
-            if (prev == (512 / 64) - NPROD - 1) {
+            if (prev == WAVES - NPROD - 1) {
                 __hip_atomic_store(&flags[6 + slot], 0, __ATOMIC_RELAXED,
                                    __HIP_MEMORY_SCOPE_WORKGROUP);
                 __hip_atomic_store(&flags[2 + slot], (int)it + 2,
@@ -316,15 +319,42 @@ extern "C" hipError_t mec_launch_fused3_encode_hh(int d, int p,
     static const bool enabled = !env || atoi(env) != 0; /* default ON */
     if (!enabled) return hipErrorNotSupported;
     if (args->shard_len % 1024 != 0) return hipErrorNotSupported;
-    dim3 blk(512);
+    /* MEC_F3_CFG: force 8- or 4-wave config (perf sweeps); MEC_F3_MIN:
+     * minimum grid (WGs) below which the pair path is used instead */
+    static const int f3cfg = [] {
+        const char *v = getenv("MEC_F3_CFG");
+        return v ? atoi(v) : 0;
+    }();
+    static const int f3min = [] {
+        const char *v = getenv("MEC_F3_MIN");
+        return v ? atoi(v) : 200;
+    }();
+    /* grid = n/G persistent workgroups (1/CU at the 8-wave config's 122
+     * KiB LDS).  Small batches under-fill the chip (the r2 canary caught
+     * 52 WGs at batch 256 running 6x slow), so: 8-wave config when it
+     * yields a near-full grid, 4-wave config (half the LDS, 2 WGs/CU)
+     * for mid batches, kernel-pair fallback below a floor. */
 #define X(D, P)                                                              \
     if (d == D && p == P) {                                                  \
-        constexpr int G = 64 / (D + P);                                      \
-        if (G < 1) return hipErrorNotSupported;                              \
-        dim3 grid((uint32_t)((args->n + G - 1) / G));                        \
-        hipLaunchKernelGGL((fused3_encode_hh_kernel<D, P, MAT_##D##_##P>),   \
-                           grid, blk, 0, stream, *args);                     \
-        return hipGetLastError();                                            \
+        constexpr int G8 = 64 / (D + P);                                     \
+        constexpr int G4 = 32 / (D + P);                                     \
+        if (G8 >= 1 && f3cfg != 4 && (args->n / G8 >= f3min ||            \
+                                       f3cfg == 8)) {                                \
+            dim3 grid((uint32_t)((args->n + G8 - 1) / G8));                  \
+            hipLaunchKernelGGL(                                              \
+                (fused3_encode_hh_kernel<D, P, MAT_##D##_##P, 8, 4>), grid,  \
+                dim3(512), 0, stream, *args);                                \
+            return hipGetLastError();                                        \
+        }                                                                    \
+        if (G4 >= 1 && f3cfg != 8 && (args->n / G4 >= f3min ||            \
+                                       f3cfg == 4)) {                                \
+            dim3 grid((uint32_t)((args->n + G4 - 1) / G4));                  \
+            hipLaunchKernelGGL(                                              \
+                (fused3_encode_hh_kernel<D, P, MAT_##D##_##P, 4, 2>), grid,  \
+                dim3(256), 0, stream, *args);                                \
+            return hipGetLastError();                                        \
+        }                                                                    \
+        return hipErrorNotSupported;                                         \
     }
     MEC_SPECIALIZED_GEOS(X)
 #undef X
