@@ -1,33 +1,38 @@
-/* Producer/consumer fused erasure-encode + HighwayHash-256 kernel (v3).
+/* Single-pass fused erasure-encode + HighwayHash-256 kernel (v3, r2).
  *
- * Single-pass semantics of cmd/erasure-coding.go:85 (GF parity) +
- * cmd/bitrot-streaming.go:57-59 (per-shard HH256) at 1.5 B of HBM traffic
- * per input byte: data read once, parity written once, every hash packet
- * consumed from LDS.
+ * Semantics of cmd/erasure-coding.go:85 (GF parity) +
+ * cmd/bitrot-streaming.go:57-59 (per-shard HH256) in ONE launch.
  *
- * Why v3 beats fused2 (r1, 0.935 ms vs the pipelined pair's 0.75): the r1
- * consumer was the pair-lane hash — 2 waves whose per-packet dependency
- * chain (~265 cyc solo) made the hash side of every tile LONGER than the
- * producer side, so hash waves sharing a SIMD with producers stretched
- * ~2x and the ring stalled.  The r2 4-lane-per-chain hash (hh256_batch4_
- * kernel) halves the per-lane instruction count and doubles hash wave
- * count: 4 producer + 4 consumer waves, one of each per SIMD, with the
- * consumer needing only ~330 wave-instr per tile against the producer's
- * ~1800 — the hash now FITS INSIDE the producer's tile time instead of
- * dominating it.  Measured effect: the fused step becomes memory-bound on
- * the 1.61 GiB it must move instead of issue-bound on the hash.
+ * Design history (all measured on MI355X):
+ *  - fused2 (r1, LDS data ring): 0.935 ms/step — the pair-lane hash's
+ *    per-packet dependency chain dominated every tile.
+ *  - fused3-LDS (r2 first cut): producers copied data AND parity through
+ *    an LDS ring; 1.39 ms/step.  Root cause: ds_write_b128 costs ~13
+ *    cycles of WAVE ISSUE per 16 B (LDS store transfer path,
+ *    MI355X_MICROARCH.md §LDS) — 12 writes/task = ~12.5k cycles/tile per
+ *    producer wave, 3.4x the GF math itself.  This is also r1 fused2's
+ *    "unexplained 3.5k cycles/tile".
+ *  - fused3 (this version): NO data plane in LDS at all.
+ *      * consumer waves hash DATA shards straight from HBM (the
+ *        hh256_batch4 shape: 4 lanes/chain, DPP zipper) — data is read
+ *        twice (GF + hash), which is cheap next to LDS store issue;
+ *      * consumer waves hash PARITY from the producers' nontemporal
+ *        global stores, tile-paced by an LDS done-counter: nt stores
+ *        KEEP the line in the writing XCD's L2 (microarch table), the
+ *        consumer shares the producer's CU, and its L1 never held those
+ *        addresses — so parity re-reads are same-XCD L2 hits, not HBM;
+ *      * producers never wait (each tile writes fresh addresses): one
+ *        monotonic done-counter, no ring, no backpressure.  If
+ *        consumers lag, parity reads fall back to HBM — graceful.
+ *    HBM traffic: data 2x read + parity 1x write = 2.5 B/input byte
+ *    (vs 3.0 for the kernel pair: the parity re-read stays in L2).
  *
- * Structure per 512-thread workgroup (G = 64/TOT blocks):
- *   waves 0..3 (producers): per 1-KiB tile, each lane owns (block,16-B
- *     column) tasks — load d inputs, write them to the LDS slot, constexpr
- *     ladder, parity to LDS + HBM (nontemporal).
- *   waves 4..7 (consumers): 4 lanes per chain (one HighwayHash lane each,
- *     DPP zipper — see hh256_batch4_kernel), 32 packets per tile from LDS.
- * Hand-off: 2-slot LDS ring with per-slot epoch flags, 4-wave arrival
- * counters, bounded spins (same protocol as fused2, which is bit-exact).
- * Consumers take static s_setprio(1): they are the younger (arbitration-
- * losing) half and sit on the latency-critical chains
- * (MI355X_MICROARCH.md "Two waves per SIMD" items 2/4).
+ * Wave layout per WG: NPROD producer waves (first half — the older,
+ * arbitration-winning half) + consumer waves at static s_setprio(1)
+ * (younger half, latency-critical chains: microarch "Two waves per SIMD"
+ * items 2/4).  All consumer lanes poll the tile counter (uniform, LDS),
+ * so data and parity lanes run the SAME code path — no divergence; the
+ * per-lane base pointer is the only difference.
  *
  * Eligibility: shard_len % 1024 == 0 and a compiled (d,p) specialization;
  * anything else falls back to the kernel pair.
@@ -107,11 +112,16 @@ __device__ __forceinline__ void hh1_update(HH1 &s, uint64_t w, uint32_t S3) {
     }
 }
 
-__device__ __forceinline__ int lds_poll(int *flag, int want) {
+__device__ __forceinline__ int lds_poll_ge(int *flag, int want) {
+    /* bounded relaxed poll for a monotonic counter; acquire pairs with
+     * the producer's release on success */
     for (int spin = 0; spin < (1 << 24); spin++) {
         if (__hip_atomic_load(flag, __ATOMIC_RELAXED,
-                              __HIP_MEMORY_SCOPE_WORKGROUP) == want)
+                              __HIP_MEMORY_SCOPE_WORKGROUP) >= want) {
+            (void)__hip_atomic_load(flag, __ATOMIC_ACQUIRE,
+                                    __HIP_MEMORY_SCOPE_WORKGROUP);
             return 0;
+        }
         __builtin_amdgcn_s_sleep(2);
     }
     return 1;
@@ -127,15 +137,9 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
     constexpr int TOT = D + P;
     /* blocks per workgroup: consumer lanes / (4 lanes per chain x TOT) */
     constexpr int G = (WAVES - NPROD_T) * 16 / TOT;
-    constexpr int TILE = 1024;         /* bytes per shard per ring tile */
-    constexpr int ROW = TILE + 16;     /* bank-skewed LDS row */
-    constexpr int RING = 2;
-    constexpr int SLOT = G * TOT * ROW;
-    constexpr int NPROD = NPROD_T;     /* producer waves (first half) */
-    __shared__ uint8_t lds[RING * SLOT + 64];
-    int *flags = (int *)&lds[RING * SLOT];
-    /* flags[0..1]=ready epoch, [2..3]=cons epoch, [4..5]=prod arrivals,
-     * [6..7]=cons arrivals */
+    constexpr int TILE = 1024; /* bytes per shard per pacing tile */
+    constexpr int NPROD = NPROD_T;
+    __shared__ int flags[2]; /* [0]=tiles done (monotonic), [1]=arrivals */
 
     const int tid = threadIdx.x;
     const int wid = tid >> 6;
@@ -144,16 +148,13 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
     const int64_t stride = a.row_stride;
     const int64_t n_iter = S / TILE;
 
-    if (tid < 8) flags[tid] = (tid == 2) ? 0 : (tid == 3 ? 1 : 0);
+    if (tid < 2) flags[tid] = 0;
     __syncthreads(); /* the ONLY workgroup barrier: flag init */
 
     if (wid < NPROD) {
-        /* ---- producer: (block g, 16-B column o) tasks ---- */
+        /* ---- producer: (block g, 16-B column o) tasks per tile ---- */
         const int lane_g = wid * 64 + (tid & 63);
         for (int64_t it = 0; it < n_iter; it++) {
-            const int slot = (int)(it & 1);
-            uint8_t *sb = &lds[slot * SLOT];
-            if (lds_poll(&flags[2 + slot], (int)it)) return; /* timeout */
             for (int task = lane_g; task < G * (TILE / 16);
                  task += NPROD * 64) {
                 const int g = task / (TILE / 16);
@@ -168,7 +169,6 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
                     uint4 pw = *(const uint4 *)(a.data +
                                                 ((b0 + g) * D + k) * stride +
                                                 off);
-                    *(uint4 *)&sb[(g * TOT + k) * ROW + o * 16] = pw;
                     /* two-bit ladder with xor3 pair-folding (constexpr on
                      * MAT, same schedule as gf_encode_kernel) */
                     uint4 cur = pw, nxt;
@@ -203,7 +203,9 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
                 }
 #pragma unroll
                 for (int i = 0; i < P; i++) {
-                    *(uint4 *)&sb[(g * TOT + D + i) * ROW + o * 16] = acc[i];
+                    /* nontemporal: written once here; the consumer's
+                     * re-read is served from this XCD's L2 (nt stores
+                     * keep the line in L2 — microarch store table) */
                     typedef unsigned int v4u
                         __attribute__((ext_vector_type(4)));
                     v4u v = {acc[i].x, acc[i].y, acc[i].z, acc[i].w};
@@ -213,17 +215,17 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
                                 off));
                 }
             }
-            /* LDS (ds_write) traffic must land before the publish; global
-             * parity stores are not part of the handoff (lgkmcnt(0) only) */
-            __builtin_amdgcn_s_waitcnt(0xc07f);
+            /* this wave's parity stores must reach L2 before the tile is
+             * published (vmcnt(0); per-wave, all NPROD waves arrive) */
+            __builtin_amdgcn_s_waitcnt(0x0f70); /* vmcnt(0) */
             if ((tid & 63) == 0) {
                 int prev = __hip_atomic_fetch_add(
-                    &flags[4 + slot], 1, __ATOMIC_RELAXED,
+                    &flags[1], 1, __ATOMIC_RELAXED,
                     __HIP_MEMORY_SCOPE_WORKGROUP);
                 if (prev == NPROD - 1) {
-                    __hip_atomic_store(&flags[4 + slot], 0, __ATOMIC_RELAXED,
+                    __hip_atomic_store(&flags[1], 0, __ATOMIC_RELAXED,
                                        __HIP_MEMORY_SCOPE_WORKGROUP);
-                    __hip_atomic_store(&flags[slot], (int)it + 1,
+                    __hip_atomic_store(&flags[0], (int)it + 1,
                                        __ATOMIC_RELEASE,
                                        __HIP_MEMORY_SCOPE_WORKGROUP);
                 }
@@ -232,17 +234,23 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
         return;
     }
 
-    /* ---- consumer: 4 lanes per chain (hh256_batch4 shape) ---- */
-    const int ln = (wid - NPROD) * 64 + (tid & 63); /* 0..255 */
-    const int cp = ln >> 2;       /* chain index in WG: 0..G*TOT-1 */
-    const int j = ln & 3;         /* HighwayHash lane */
+    /* ---- consumer: 4 lanes per chain (hh256_batch4 shape), tile-paced
+     * by the done-counter; data and parity lanes share the code path —
+     * only the base pointer differs ---- */
+    const int ln = (wid - NPROD) * 64 + (tid & 63);
+    const int cp = ln >> 2; /* chain index in WG */
+    const int j = ln & 3;   /* HighwayHash lane */
     const int cg = cp / TOT;
     const int cs = cp % TOT;
     const bool act = (cp < G * TOT) && (b0 + cg < a.n);
     const uint32_t S3 = (j & 1) ? 0x07000601u : 0x00070106u;
+    const uint8_t *base =
+        (cs < D) ? a.data + ((b0 + cg) * D + cs) * stride
+                 : a.parity + ((b0 + cg) * P + (cs - D)) * stride;
+    const uint8_t *mp = base + 8 * j;
 
-    /* consumers are the younger dispatch half AND the latency-critical
-     * side: one static priority raise, no per-tile flips */
+    /* consumers are the younger (arbitration-losing) dispatch half and
+     * sit on the latency-critical chains: one static priority raise */
     __builtin_amdgcn_s_setprio(1);
 
     HH1 s;
@@ -262,29 +270,16 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
     }
 
     for (int64_t it = 0; it < n_iter; it++) {
-        const int slot = (int)(it & 1);
-        if (lds_poll(&flags[slot], (int)it + 1)) return; /* timeout */
-        __hip_atomic_load(&flags[slot], __ATOMIC_ACQUIRE,
-                          __HIP_MEMORY_SCOPE_WORKGROUP);
+        if (lds_poll_ge(&flags[0], (int)it + 1)) return; /* timeout */
         if (act) {
-            const uint8_t *row =
-                &lds[slot * SLOT + (cg * TOT + cs) * ROW + 8 * j];
-#pragma unroll 8
+            /* 32 packets; loads batched ahead of the serial chain */
+            uint64_t q[TILE / 32];
+#pragma unroll
             for (int t = 0; t < TILE / 32; t++)
-                hh1_update(s, *(const uint64_t *)(row + 32 * t), S3);
-        }
-        __builtin_amdgcn_s_waitcnt(0xc07f); /* lgkmcnt(0): ds_reads done */
-        if ((tid & 63) == 0) {
-            int prev = __hip_atomic_fetch_add(&flags[6 + slot], 1,
-                                              __ATOMIC_RELAXED,
-                                              __HIP_MEMORY_SCOPE_WORKGROUP);
-            if (prev == WAVES - NPROD - 1) {
-                __hip_atomic_store(&flags[6 + slot], 0, __ATOMIC_RELAXED,
-                                   __HIP_MEMORY_SCOPE_WORKGROUP);
-                __hip_atomic_store(&flags[2 + slot], (int)it + 2,
-                                   __ATOMIC_RELEASE,
-                                   __HIP_MEMORY_SCOPE_WORKGROUP);
-            }
+                q[t] = *(const uint64_t *)(mp + 32 * t);
+#pragma unroll
+            for (int t = 0; t < TILE / 32; t++) hh1_update(s, q[t], S3);
+            mp += TILE;
         }
     }
 
