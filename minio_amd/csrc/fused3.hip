@@ -112,13 +112,23 @@ __device__ __forceinline__ void hh1_update(HH1 &s, uint64_t w, uint32_t S3) {
     }
 }
 
-__device__ __forceinline__ int lds_poll_ge(int *flag, int want) {
-    /* bounded relaxed poll for a monotonic counter; acquire pairs with
-     * the producer's release on success */
+template <int NPROD>
+__device__ __forceinline__ int lds_poll_min_ge(int *prog, int want) {
+    /* bounded relaxed poll over NPROD per-wave monotonic progress
+     * counters (no shared arrival counter: producer waves never wait, so
+     * a conflated counter publishes early/loses publishes when waves run
+     * ahead of each other — r2 bug).  Acquire pairs with each wave's
+     * release on success. */
     for (int spin = 0; spin < (1 << 24); spin++) {
-        if (__hip_atomic_load(flag, __ATOMIC_RELAXED,
-                              __HIP_MEMORY_SCOPE_WORKGROUP) >= want) {
-            (void)__hip_atomic_load(flag, __ATOMIC_ACQUIRE,
+        int mn = 1 << 30;
+#pragma unroll
+        for (int w = 0; w < NPROD; w++) {
+            int v = __hip_atomic_load(&prog[w], __ATOMIC_RELAXED,
+                                      __HIP_MEMORY_SCOPE_WORKGROUP);
+            mn = v < mn ? v : mn;
+        }
+        if (mn >= want) {
+            (void)__hip_atomic_load(&prog[0], __ATOMIC_ACQUIRE,
                                     __HIP_MEMORY_SCOPE_WORKGROUP);
             return 0;
         }
@@ -139,7 +149,7 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
     constexpr int G = (WAVES - NPROD_T) * 16 / TOT;
     constexpr int TILE = 1024; /* bytes per shard per pacing tile */
     constexpr int NPROD = NPROD_T;
-    __shared__ int flags[2]; /* [0]=tiles done (monotonic), [1]=arrivals */
+    __shared__ int prog[NPROD_T]; /* per-producer-wave tiles-done */
 
     const int tid = threadIdx.x;
     const int wid = tid >> 6;
@@ -148,8 +158,8 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
     const int64_t stride = a.row_stride;
     const int64_t n_iter = S / TILE;
 
-    if (tid < 2) flags[tid] = 0;
-    __syncthreads(); /* the ONLY workgroup barrier: flag init */
+    if (tid < NPROD_T) prog[tid] = 0;
+    __syncthreads(); /* the ONLY workgroup barrier: counter init */
 
     if (wid < NPROD) {
         /* ---- producer: (block g, 16-B column o) tasks per tile ---- */
@@ -215,21 +225,14 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
                                 off));
                 }
             }
-            /* this wave's parity stores must reach L2 before the tile is
-             * published (vmcnt(0); per-wave, all NPROD waves arrive) */
+            /* this wave's parity stores must reach L2 before its tile
+             * progress is published (vmcnt(0), then a released per-wave
+             * store — no cross-wave counter, so no waiting) */
             __builtin_amdgcn_s_waitcnt(0x0f70); /* vmcnt(0) */
-            if ((tid & 63) == 0) {
-                int prev = __hip_atomic_fetch_add(
-                    &flags[1], 1, __ATOMIC_RELAXED,
-                    __HIP_MEMORY_SCOPE_WORKGROUP);
-                if (prev == NPROD - 1) {
-                    __hip_atomic_store(&flags[1], 0, __ATOMIC_RELAXED,
-                                       __HIP_MEMORY_SCOPE_WORKGROUP);
-                    __hip_atomic_store(&flags[0], (int)it + 1,
-                                       __ATOMIC_RELEASE,
-                                       __HIP_MEMORY_SCOPE_WORKGROUP);
-                }
-            }
+            if ((tid & 63) == 0)
+                __hip_atomic_store(&prog[wid], (int)it + 1,
+                                   __ATOMIC_RELEASE,
+                                   __HIP_MEMORY_SCOPE_WORKGROUP);
         }
         return;
     }
@@ -270,7 +273,7 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
     }
 
     for (int64_t it = 0; it < n_iter; it++) {
-        if (lds_poll_ge(&flags[0], (int)it + 1)) return; /* timeout */
+        if (lds_poll_min_ge<NPROD>(prog, (int)it + 1)) return; /* timeout */
         if (act) {
             /* 32 packets; loads batched ahead of the serial chain */
             uint64_t q[TILE / 32];

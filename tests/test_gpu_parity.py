@@ -591,7 +591,8 @@ print("KNOB_OK")
 
 @pytest.mark.parametrize("knob", ["MEC_HH_LDS", "MEC_FUSED2", "MEC_FUSED",
                                   "MEC_HH_WG", "MEC_HH4_OFF", "MEC_SHA_WG",
-                                  "MEC_FUSED3_OFF"])
+                                  "MEC_FUSED3_OFF", "MEC_F3_MIN1",
+                                  "MEC_F3_MIN1_W4"])
 def test_knob_variants_bit_exact(knob):
     """The in-tree experiment knobs are env-latched at first
     use, so each variant runs in a subprocess.  Every knob'd kernel must
@@ -605,6 +606,12 @@ def test_knob_variants_bit_exact(knob):
         env["MEC_HH4"] = "0"   # the r1 pair-lane kernel
     elif knob == "MEC_FUSED3_OFF":
         env["MEC_FUSED3"] = "0"  # two-kernel pair instead of fused v3
+    elif knob == "MEC_F3_MIN1":
+        env["MEC_F3_MIN"] = "1"  # force fused3 8-wave at ANY batch size
+        env["MEC_F3_CFG"] = "8"
+    elif knob == "MEC_F3_MIN1_W4":
+        env["MEC_F3_MIN"] = "1"  # force fused3 4-wave at ANY batch size
+        env["MEC_F3_CFG"] = "4"
     else:
         env[knob] = "1"
     env["MEC_TEST_REPO"] = os.path.dirname(HERE)
