@@ -162,8 +162,96 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
     __syncthreads(); /* the ONLY workgroup barrier: counter init */
 
     if (wid < NPROD) {
-        /* ---- producer: (block g, 16-B column o) tasks per tile ---- */
+        /* ---- producer: (block g, 16-B column o) tasks per tile ----
+         * All of a task's D row loads are issued before any ladder math
+         * (the consume-as-you-load form exposes one HBM latency per row
+         * — r03 PMC; r2 profile showed 49% of fused wave-cycles parked
+         * for exactly this reason), and when each lane owns exactly one
+         * task (EXACT), the NEXT tile's loads are issued before the
+         * current tile's ladder so the latency rides under compute. */
+        constexpr bool EXACT = (G * (TILE / 16) == NPROD * 64);
         const int lane_g = wid * 64 + (tid & 63);
+        if (EXACT) {
+            const int g = lane_g / (TILE / 16);
+            const int o = lane_g % (TILE / 16);
+            const bool pact = b0 + g < a.n;
+            const uint8_t *rows[D];
+#pragma unroll
+            for (int k = 0; k < D; k++)
+                rows[k] = a.data + ((b0 + g) * D + k) * stride +
+                          (int64_t)o * 16;
+            uint4 pwsA[D], pwsB[D];
+            if (pact) {
+#pragma unroll
+                for (int k = 0; k < D; k++) pwsA[k] = *(const uint4 *)rows[k];
+            }
+            for (int64_t it = 0; it < n_iter; it++) {
+                if (pact && it + 1 < n_iter) {
+#pragma unroll
+                    for (int k = 0; k < D; k++)
+                        pwsB[k] =
+                            *(const uint4 *)(rows[k] + (it + 1) * TILE);
+                }
+                if (pact) {
+                    const int64_t off = it * TILE + (int64_t)o * 16;
+                    uint4 acc[P];
+#pragma unroll
+                    for (int i = 0; i < P; i++) acc[i] = uint4{0, 0, 0, 0};
+#pragma unroll
+                    for (int k = 0; k < D; k++) {
+                        uint4 cur = pwsA[k], nxt;
+#pragma unroll
+                        for (int bit = 0; bit < 8; bit += 2) {
+                            uint32_t needCur = 0, needHi = 0;
+#pragma unroll
+                            for (int i = 0; i < P; i++) {
+                                needCur |= (uint32_t)MAT[i][k] >> bit;
+                                needHi |= (uint32_t)MAT[i][k] >> (bit + 1);
+                            }
+                            if (!needCur) break;
+                            if (needHi) {
+                                nxt = cur;
+                                gf2x4(nxt);
+                            }
+#pragma unroll
+                            for (int i = 0; i < P; i++) {
+                                const int b0i = (MAT[i][k] >> bit) & 1;
+                                const int b1i = (MAT[i][k] >> (bit + 1)) & 1;
+                                if (b0i && b1i) xor34(acc[i], cur, nxt);
+                                else if (b0i) xor4(acc[i], cur);
+                                else if (b1i) xor4(acc[i], nxt);
+                            }
+                            if (needHi >> 1) {
+                                cur = nxt;
+                                gf2x4(cur);
+                            } else {
+                                break;
+                            }
+                        }
+                    }
+#pragma unroll
+                    for (int i = 0; i < P; i++) {
+                        typedef unsigned int v4u
+                            __attribute__((ext_vector_type(4)));
+                        v4u v = {acc[i].x, acc[i].y, acc[i].z, acc[i].w};
+                        __builtin_nontemporal_store(
+                            v, (v4u *)(a.parity +
+                                       ((b0 + g) * P + i) * stride + off));
+                    }
+#pragma unroll
+                    for (int k = 0; k < D; k++) pwsA[k] = pwsB[k];
+                }
+                /* parity stores to L2 before publishing this tile */
+                __builtin_amdgcn_s_waitcnt(0x0f70); /* vmcnt(0) */
+                if ((tid & 63) == 0)
+                    __hip_atomic_store(&prog[wid], (int)it + 1,
+                                       __ATOMIC_RELEASE,
+                                       __HIP_MEMORY_SCOPE_WORKGROUP);
+            }
+            return;
+        }
+        /* generic task loop (lane count != task count): loads hoisted
+         * per task, no cross-tile prefetch */
         for (int64_t it = 0; it < n_iter; it++) {
             for (int task = lane_g; task < G * (TILE / 16);
                  task += NPROD * 64) {
@@ -171,17 +259,18 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
                 const int o = task % (TILE / 16);
                 if (b0 + g >= a.n) continue;
                 const int64_t off = it * TILE + (int64_t)o * 16;
+                uint4 pws[D];
+#pragma unroll
+                for (int k = 0; k < D; k++)
+                    pws[k] = *(const uint4 *)(a.data +
+                                              ((b0 + g) * D + k) * stride +
+                                              off);
                 uint4 acc[P];
 #pragma unroll
                 for (int i = 0; i < P; i++) acc[i] = uint4{0, 0, 0, 0};
 #pragma unroll
                 for (int k = 0; k < D; k++) {
-                    uint4 pw = *(const uint4 *)(a.data +
-                                                ((b0 + g) * D + k) * stride +
-                                                off);
-                    /* two-bit ladder with xor3 pair-folding (constexpr on
-                     * MAT, same schedule as gf_encode_kernel) */
-                    uint4 cur = pw, nxt;
+                    uint4 cur = pws[k], nxt;
 #pragma unroll
                     for (int bit = 0; bit < 8; bit += 2) {
                         uint32_t needCur = 0, needHi = 0;
@@ -213,9 +302,6 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
                 }
 #pragma unroll
                 for (int i = 0; i < P; i++) {
-                    /* nontemporal: written once here; the consumer's
-                     * re-read is served from this XCD's L2 (nt stores
-                     * keep the line in L2 — microarch store table) */
                     typedef unsigned int v4u
                         __attribute__((ext_vector_type(4)));
                     v4u v = {acc[i].x, acc[i].y, acc[i].z, acc[i].w};
@@ -225,9 +311,6 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
                                 off));
                 }
             }
-            /* this wave's parity stores must reach L2 before its tile
-             * progress is published (vmcnt(0), then a released per-wave
-             * store — no cross-wave counter, so no waiting) */
             __builtin_amdgcn_s_waitcnt(0x0f70); /* vmcnt(0) */
             if ((tid & 63) == 0)
                 __hip_atomic_store(&prog[wid], (int)it + 1,
@@ -272,17 +355,38 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
         s.v1 = init1[j] ^ ((a.key[j] >> 32) | (a.key[j] << 32));
     }
 
-    for (int64_t it = 0; it < n_iter; it++) {
-        if (lds_poll_min_ge<NPROD>(prog, (int)it + 1)) return; /* timeout */
-        if (act) {
-            /* 32 packets; loads batched ahead of the serial chain */
-            uint64_t q[TILE / 32];
+    /* run ONE TILE BEHIND the producers: tile t+1's 32 loads are issued
+     * before tile t's serial chain, so the (L2-or-HBM) load latency rides
+     * under ~1.7k cycles of hashing instead of stalling every tile */
+    {
+        uint64_t qA[TILE / 32], qB[TILE / 32];
+        if (n_iter > 0) {
+            if (lds_poll_min_ge<NPROD>(prog, 1)) return;
+            if (act) {
 #pragma unroll
-            for (int t = 0; t < TILE / 32; t++)
-                q[t] = *(const uint64_t *)(mp + 32 * t);
+                for (int t = 0; t < TILE / 32; t++)
+                    qA[t] = *(const uint64_t *)(mp + 32 * t);
+                mp += TILE;
+            }
+        }
+        for (int64_t it = 0; it < n_iter; it++) {
+            if (it + 1 < n_iter) {
+                if (lds_poll_min_ge<NPROD>(prog, (int)it + 2))
+                    return; /* timeout */
+                if (act) {
 #pragma unroll
-            for (int t = 0; t < TILE / 32; t++) hh1_update(s, q[t], S3);
-            mp += TILE;
+                    for (int t = 0; t < TILE / 32; t++)
+                        qB[t] = *(const uint64_t *)(mp + 32 * t);
+                    mp += TILE;
+                }
+            }
+            if (act) {
+#pragma unroll
+                for (int t = 0; t < TILE / 32; t++)
+                    hh1_update(s, qA[t], S3);
+#pragma unroll
+                for (int t = 0; t < TILE / 32; t++) qA[t] = qB[t];
+            }
         }
     }
 
