@@ -160,6 +160,7 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
 
     if (tid < NPROD_T) prog[tid] = 0;
     __syncthreads(); /* the ONLY workgroup barrier: counter init */
+    if (a.probe == 1 && wid >= NPROD_T) return;
 
     if (wid < NPROD) {
         /* ---- producer: (block g, 16-B column o) tasks per tile ----
@@ -361,7 +362,7 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
     {
         uint64_t qA[TILE / 32], qB[TILE / 32];
         if (n_iter > 0) {
-            if (lds_poll_min_ge<NPROD>(prog, 1)) return;
+            if (a.probe < 2 && lds_poll_min_ge<NPROD>(prog, 1)) return;
             if (act) {
 #pragma unroll
                 for (int t = 0; t < TILE / 32; t++)
@@ -371,7 +372,8 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
         }
         for (int64_t it = 0; it < n_iter; it++) {
             if (it + 1 < n_iter) {
-                if (lds_poll_min_ge<NPROD>(prog, (int)it + 2))
+                if (a.probe < 2 &&
+                    lds_poll_min_ge<NPROD>(prog, (int)it + 2))
                     return; /* timeout */
                 if (act) {
 #pragma unroll
@@ -431,6 +433,13 @@ extern "C" hipError_t mec_launch_fused3_encode_hh(int d, int p,
         const char *v = getenv("MEC_F3_MIN");
         return v ? atoi(v) : 200;
     }();
+    static const int f3probe = [] {
+        const char *v = getenv("MEC_F3_PROBE");
+        return v ? atoi(v) : 0;
+    }();
+    FusedArgs aa = *args;
+    aa.probe = f3probe;
+    args = &aa;
     /* grid = n/G persistent workgroups (1/CU at the 8-wave config's 122
      * KiB LDS).  Small batches under-fill the chip (the r2 canary caught
      * 52 WGs at batch 256 running 6x slow), so: 8-wave config when it

@@ -61,6 +61,9 @@ struct FusedArgs {
     int64_t shard_len;
     int64_t n;
     uint64_t key[4];
+    int probe; /* 0 normal; perf-isolation probes (results INVALID):
+                  1 producers only, 2 consumers free-run (no poll),
+                  3 both free-run */
 };
 
 struct ScatterArgs {
