@@ -148,6 +148,11 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
     /* blocks per workgroup: consumer lanes / (4 lanes per chain x TOT) */
     constexpr int G = (WAVES - NPROD_T) * 16 / TOT;
     constexpr int TILE = 1024; /* bytes per shard per pacing tile */
+    constexpr int PUBK = 8;    /* publish cadence (tiles): a vmcnt(0)
+        store-drain costs ~1-3 us under full-chip load (microarch
+        publish-large row) — draining EVERY tile was 0.62 ms of the r2
+        fused step (probe1).  Publishing every PUBK tiles cuts that 8x;
+        consumers simply run up to PUBK tiles behind. */
     constexpr int NPROD = NPROD_T;
     __shared__ int prog[NPROD_T]; /* per-producer-wave tiles-done */
 
@@ -242,12 +247,14 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
 #pragma unroll
                     for (int k = 0; k < D; k++) pwsA[k] = pwsB[k];
                 }
-                /* parity stores to L2 before publishing this tile */
-                __builtin_amdgcn_s_waitcnt(0x0f70); /* vmcnt(0) */
-                if ((tid & 63) == 0)
-                    __hip_atomic_store(&prog[wid], (int)it + 1,
-                                       __ATOMIC_RELEASE,
-                                       __HIP_MEMORY_SCOPE_WORKGROUP);
+                /* parity stores to L2 before publishing (PUBK cadence) */
+                if ((it + 1) % PUBK == 0 || it + 1 == n_iter) {
+                    __builtin_amdgcn_s_waitcnt(0x0f70); /* vmcnt(0) */
+                    if ((tid & 63) == 0)
+                        __hip_atomic_store(&prog[wid], (int)it + 1,
+                                           __ATOMIC_RELEASE,
+                                           __HIP_MEMORY_SCOPE_WORKGROUP);
+                }
             }
             return;
         }
@@ -312,11 +319,13 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
                                 off));
                 }
             }
-            __builtin_amdgcn_s_waitcnt(0x0f70); /* vmcnt(0) */
-            if ((tid & 63) == 0)
-                __hip_atomic_store(&prog[wid], (int)it + 1,
-                                   __ATOMIC_RELEASE,
-                                   __HIP_MEMORY_SCOPE_WORKGROUP);
+            if ((it + 1) % PUBK == 0 || it + 1 == n_iter) {
+                __builtin_amdgcn_s_waitcnt(0x0f70); /* vmcnt(0) */
+                if ((tid & 63) == 0)
+                    __hip_atomic_store(&prog[wid], (int)it + 1,
+                                       __ATOMIC_RELEASE,
+                                       __HIP_MEMORY_SCOPE_WORKGROUP);
+            }
         }
         return;
     }
