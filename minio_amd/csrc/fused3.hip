@@ -145,8 +145,16 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
     FusedArgs a) {
     using namespace fused3;
     constexpr int TOT = D + P;
-    /* blocks per workgroup: consumer lanes / (4 lanes per chain x TOT) */
-    constexpr int G = (WAVES - NPROD_T) * 16 / TOT;
+    /* blocks per workgroup: G = NPROD makes the producer side EXACTLY one
+     * 16-B task per lane per tile (and, at the headline batch, a grid
+     * that covers all 256 CUs), capped by consumer lanes (4 per chain).
+     * The 8-wave config's cyclic wave->SIMD placement then puts exactly
+     * one producer + one consumer wave on every SIMD (waves i and i+4
+     * share a SIMD) — deterministic balance; the 4-wave config relies on
+     * the dispatcher mixing two WGs per CU. */
+    constexpr int G = NPROD_T < (WAVES - NPROD_T) * 16 / TOT
+                          ? NPROD_T
+                          : (WAVES - NPROD_T) * 16 / TOT;
     constexpr int TILE = 1024; /* bytes per shard per pacing tile */
     constexpr int PUBK = 8;    /* publish cadence (tiles): a vmcnt(0)
         store-drain costs ~1-3 us under full-chip load (microarch
