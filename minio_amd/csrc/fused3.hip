@@ -464,8 +464,8 @@ extern "C" hipError_t mec_launch_fused3_encode_hh(int d, int p,
      * for mid batches, kernel-pair fallback below a floor. */
 #define X(D, P)                                                              \
     if (d == D && p == P) {                                                  \
-        constexpr int G8 = 64 / (D + P);                                     \
-        constexpr int G4 = 32 / (D + P);                                     \
+        constexpr int G8 = 4 < 64 / (D + P) ? 4 : 64 / (D + P);             \
+        constexpr int G4 = 2 < 32 / (D + P) ? 2 : 32 / (D + P);             \
         if (G8 >= 1 && f3cfg != 4 && (args->n / G8 >= f3min ||            \
                                        f3cfg == 8)) {                                \
             dim3 grid((uint32_t)((args->n + G8 - 1) / G8));                  \
