@@ -173,7 +173,8 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
 
     if (tid < NPROD_T) prog[tid] = 0;
     __syncthreads(); /* the ONLY workgroup barrier: counter init */
-    if (a.probe == 1 && wid >= NPROD_T) return;
+    if (a.probe >= 4) { if (wid >= NPROD_T) return; }
+    else if (a.probe == 1 && wid >= NPROD_T) return;
 
     if (wid < NPROD) {
         /* ---- producer: (block g, 16-B column o) tasks per tile ----
@@ -256,7 +257,8 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
                     for (int k = 0; k < D; k++) pwsA[k] = pwsB[k];
                 }
                 /* parity stores to L2 before publishing (PUBK cadence) */
-                if ((it + 1) % PUBK == 0 || it + 1 == n_iter) {
+                if (a.probe != 4 &&
+                    ((it + 1) % PUBK == 0 || it + 1 == n_iter)) {
                     __builtin_amdgcn_s_waitcnt(0x0f70); /* vmcnt(0) */
                     if ((tid & 63) == 0)
                         __hip_atomic_store(&prog[wid], (int)it + 1,
