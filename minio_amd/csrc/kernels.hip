@@ -183,6 +183,172 @@ __global__ void __launch_bounds__(256) gf_encode_kernel(GfEncArgs a) {
     }
 }
 
+/* ---- bit-sliced specialized encode (r2) --------------------------------
+ *
+ * The SWAR xtime ladder costs ~1.46 VALU per input byte (and the int-VALU
+ * pipe measures ~4 cyc/instr on gfx950, making the GF leg VALU-PIPE-bound
+ * at ~0.32 ms for the headline batch — r2 SQ counters).  Bit-slicing cuts
+ * the instruction count ~2.4x: each lane takes a 32-byte column per row,
+ * transposes it to 8 bit-planes (3-stage delta-swap network, an
+ * involution — verified exhaustively on CPU), and then EVERY GF(2^8)
+ * constant-multiply-accumulate is a straight-line XOR of planes with the
+ * coefficient's bit-matrix rows folded at compile time (constexpr over
+ * MAT): ~16 xors per (input,parity) pair instead of the ladder's ~90
+ * slots.  Transposes: 60 ops per 32 B, amortized over P outputs.
+ */
+__device__ __forceinline__ void bs_pair(uint32_t &A, uint32_t &B, int s,
+                                        uint32_t m) {
+    /* delta-swap: exchanges bit groups between regs A and B */
+    uint32_t t = (uint32_t)__builtin_amdgcn_bitop3_b32(B << s, A, m,
+                                                       0x28); /* (b^a)&m */
+    A ^= t;
+    B ^= t >> s;
+}
+
+__device__ __forceinline__ void bs_transpose(uint32_t r[8]) {
+    /* stages d=1,2,4: after this, r[b] holds bit b of all 32 bytes
+     * (slot order is a fixed byte permutation, identical across planes;
+     * the same network inverts it — involution) */
+    bs_pair(r[0], r[1], 1, 0xAAAAAAAAu);
+    bs_pair(r[2], r[3], 1, 0xAAAAAAAAu);
+    bs_pair(r[4], r[5], 1, 0xAAAAAAAAu);
+    bs_pair(r[6], r[7], 1, 0xAAAAAAAAu);
+    bs_pair(r[0], r[2], 2, 0xCCCCCCCCu);
+    bs_pair(r[1], r[3], 2, 0xCCCCCCCCu);
+    bs_pair(r[4], r[6], 2, 0xCCCCCCCCu);
+    bs_pair(r[5], r[7], 2, 0xCCCCCCCCu);
+    bs_pair(r[0], r[4], 4, 0xF0F0F0F0u);
+    bs_pair(r[1], r[5], 4, 0xF0F0F0F0u);
+    bs_pair(r[2], r[6], 4, 0xF0F0F0F0u);
+    bs_pair(r[3], r[7], 4, 0xF0F0F0F0u);
+}
+
+/* constexpr GF(2^8)/0x11D multiply and bit-matrix row masks */
+constexpr uint8_t bs_gfmul(uint8_t a, uint8_t b) {
+    uint32_t r = 0, x = a;
+    for (int i = 0; i < 8; i++) {
+        if ((b >> i) & 1) r ^= x << i;
+    }
+    /* reduce 15-bit poly product mod 0x11D */
+    for (int i = 14; i >= 8; i--)
+        if ((r >> i) & 1) r ^= 0x11Du << (i - 8);
+    return (uint8_t)r;
+}
+/* rowmask(c, b) bit a: output bit b of c*x depends on input bit a */
+constexpr uint8_t bs_rowmask(uint8_t c, int b) {
+    uint8_t m = 0;
+    for (int a = 0; a < 8; a++)
+        if ((bs_gfmul(c, (uint8_t)(1u << a)) >> b) & 1)
+            m |= (uint8_t)(1u << a);
+    return m;
+}
+
+/* explicit xor3 pair-folding of a plane-XOR set (the compiler leaves
+ * these as chains of v_xor otherwise — measured 1267 plain xors/loop) */
+template <uint8_t M>
+__device__ __forceinline__ uint32_t bs_fold(const uint32_t x[8],
+                                            uint32_t acc) {
+    if constexpr (M == 0) {
+        return acc;
+    } else {
+        constexpr int a0 = __builtin_ctz(M);
+        constexpr uint8_t M1 = M & (M - 1);
+        if constexpr (M1 == 0) {
+            return acc ^ x[a0];
+        } else {
+            constexpr int a1 = __builtin_ctz(M1);
+            constexpr uint8_t M2 = M1 & (M1 - 1);
+            return bs_fold<M2>(
+                x, (uint32_t)__builtin_amdgcn_bitop3_b32(acc, x[a0], x[a1],
+                                                         0x96));
+        }
+    }
+}
+
+/* compile-time iteration over (parity row, plane) so the fold masks are
+ * constant expressions */
+template <int D, int P, const uint8_t (&MAT)[P][D], int K, int I, int PB>
+__device__ __forceinline__ void bs_acc_all(const uint32_t xc[8],
+                                           uint32_t accp[P][8]) {
+    if constexpr (I < P) {
+        accp[I][PB] = bs_fold<bs_rowmask(MAT[I][K], PB)>(xc, accp[I][PB]);
+        if constexpr (PB < 7)
+            bs_acc_all<D, P, MAT, K, I, PB + 1>(xc, accp);
+        else
+            bs_acc_all<D, P, MAT, K, I + 1, 0>(xc, accp);
+    }
+}
+
+template <int D, int P, const uint8_t (&MAT)[P][D], int K = 0>
+__device__ __forceinline__ void bs_acc_k(int k, const uint32_t xc[8],
+                                         uint32_t accp[P][8]) {
+    if constexpr (K < D) {
+        if (k == K)
+            bs_acc_all<D, P, MAT, K, 0, 0>(xc, accp);
+        else
+            bs_acc_k<D, P, MAT, K + 1>(k, xc, accp);
+    }
+}
+
+template <int D, int P, const uint8_t (&MAT)[P][D], bool NT>
+__global__ void __launch_bounds__(256) gf_encode_bs_kernel(GfEncArgs a) {
+    const int b = blockIdx.y;
+    const int64_t cols = (a.shard_len + 31) / 32;
+    const uint8_t *__restrict__ sbase = a.data + (int64_t)b * D * a.row_stride;
+    uint8_t *__restrict__ obase = a.parity + (int64_t)b * P * a.row_stride;
+
+    for (int64_t c = blockIdx.x * blockDim.x + threadIdx.x; c < cols;
+         c += (int64_t)gridDim.x * blockDim.x) {
+        const int64_t j = c * 32;
+        uint32_t accp[P][8];
+#pragma unroll
+        for (int i = 0; i < P; i++)
+#pragma unroll
+            for (int pb = 0; pb < 8; pb++) accp[i][pb] = 0;
+        /* software pipeline: row k+1's loads in flight during row k's
+         * transpose + plane xors */
+        uint32_t xc[8], xn[8];
+        {
+            const uint8_t *row = sbase + j;
+            uint4 lo = *(const uint4 *)row;
+            uint4 hi = *(const uint4 *)(row + 16);
+            xc[0] = lo.x; xc[1] = lo.y; xc[2] = lo.z; xc[3] = lo.w;
+            xc[4] = hi.x; xc[5] = hi.y; xc[6] = hi.z; xc[7] = hi.w;
+        }
+#pragma unroll
+        for (int k = 0; k < D; k++) {
+            if (k + 1 < D) {
+                const uint8_t *row = sbase + (int64_t)(k + 1) * a.row_stride + j;
+                uint4 lo = *(const uint4 *)row;
+                uint4 hi = *(const uint4 *)(row + 16);
+                xn[0] = lo.x; xn[1] = lo.y; xn[2] = lo.z; xn[3] = lo.w;
+                xn[4] = hi.x; xn[5] = hi.y; xn[6] = hi.z; xn[7] = hi.w;
+            }
+            bs_transpose(xc);
+            bs_acc_k<D, P, MAT>(k, xc, accp);
+#pragma unroll
+            for (int w = 0; w < 8; w++) xc[w] = xn[w];
+        }
+#pragma unroll
+        for (int i = 0; i < P; i++) {
+            bs_transpose(accp[i]);
+            uint8_t *orow = obase + (int64_t)i * a.row_stride + j;
+            uint4 lo{accp[i][0], accp[i][1], accp[i][2], accp[i][3]};
+            uint4 hi{accp[i][4], accp[i][5], accp[i][6], accp[i][7]};
+            if (NT) {
+                typedef unsigned int v4u __attribute__((ext_vector_type(4)));
+                v4u vlo = {lo.x, lo.y, lo.z, lo.w};
+                v4u vhi = {hi.x, hi.y, hi.z, hi.w};
+                __builtin_nontemporal_store(vlo, (v4u *)orow);
+                __builtin_nontemporal_store(vhi, (v4u *)(orow + 16));
+            } else {
+                *(uint4 *)orow = lo;
+                *(uint4 *)(orow + 16) = hi;
+            }
+        }
+    }
+}
+
 /* ---- generic GF matrix-multiply over shard rows ------------------------
  *
  * out[t][j] = sum_k mat[t][k] * src[k][j]  (GF(2^8)), per batch item.
@@ -1217,6 +1383,27 @@ static int gf_env_int(const char *name, int dflt) {
 
 hipError_t mec_launch_gf_encode_spec(int d, int p, const GfEncArgs *args,
                                      int n, hipStream_t stream) {
+    static const int env_bs = gf_env_int("MEC_GF_BS", 1);
+    if (env_bs) {
+        /* bit-sliced encode (r2 default): ~2.4x fewer VALU slots than the
+         * xtime ladder; 32 B per lane */
+        const int64_t cols = (args->shard_len + 31) / 32;
+        int64_t max_x = (cols + 255) / 256;
+        int64_t want_x = ((int64_t)2048 * 4 + n - 1) / n;
+        int64_t blocks_x = want_x < max_x ? want_x : max_x;
+        if (blocks_x < 1) blocks_x = 1;
+        dim3 grid((uint32_t)blocks_x, n);
+        dim3 blk(256);
+#define XBS(D, P)                                                            \
+        if (d == D && p == P) {                                              \
+            hipLaunchKernelGGL((gf_encode_bs_kernel<D, P, MAT_##D##_##P,     \
+                                                    true>),                  \
+                               grid, blk, 0, stream, *args);                 \
+            return hipGetLastError();                                        \
+        }
+        MEC_SPECIALIZED_GEOS(XBS)
+#undef XBS
+    }
     static const int env_w = gf_env_int("MEC_GF_W", 1);
     static const int env_nt = gf_env_int("MEC_GF_NT", 1);
     static const int env_wgx = gf_env_int("MEC_GF_WGX", 4);
