@@ -43,6 +43,7 @@
 
 #include "kernels.h"
 #include "ec_matrices_gen.h"
+#include "gf_bs.h"
 
 namespace fused3 {
 
@@ -140,7 +141,7 @@ __device__ __forceinline__ int lds_poll_min_ge(int *prog, int want) {
 } // namespace fused3
 
 template <int D, int P, const uint8_t (&MAT)[P][D], int WAVES = 8,
-          int NPROD_T = 4>
+          int NPROD_T = 4, int PRODBS = 0>
 __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
     FusedArgs a) {
     using namespace fused3;
@@ -152,9 +153,11 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
      * one producer + one consumer wave on every SIMD (waves i and i+4
      * share a SIMD) — deterministic balance; the 4-wave config relies on
      * the dispatcher mixing two WGs per CU. */
-    constexpr int G = NPROD_T < (WAVES - NPROD_T) * 16 / TOT
-                          ? NPROD_T
-                          : (WAVES - NPROD_T) * 16 / TOT;
+    constexpr int GCAP = (WAVES - NPROD_T) * 16 / TOT; /* consumer lanes */
+    /* byte-ladder producer: 16-B tasks, G = NPROD lanes-exact;
+     * bit-sliced producer: 32-B tasks, G = 2*NPROD lanes-exact */
+    constexpr int GWANT = PRODBS ? 2 * NPROD_T : NPROD_T;
+    constexpr int G = GWANT < GCAP ? GWANT : GCAP;
     constexpr int TILE = 1024; /* bytes per shard per pacing tile */
     constexpr int PUBK = 8;    /* publish cadence (tiles): a vmcnt(0)
         store-drain costs ~1-3 us under full-chip load (microarch
@@ -176,6 +179,74 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
     if (a.probe >= 4) { if (wid >= NPROD_T) return; }
     else if (a.probe == 1 && wid >= NPROD_T) return;
 
+    if (PRODBS && wid < NPROD) {
+        /* ---- bit-sliced producer: (block g, 32-B column o) tasks ----
+         * Same plane-transpose + constexpr xor3 network as
+         * gf_encode_bs_kernel (gf_bs.h): ~2.2x fewer VALU slots than the
+         * ladder, which matters because the fused kernel is VALU-pipe-
+         * bound (r2 SQ counters: int VALU ~4 cyc/instr). */
+        constexpr int TASKS = G * (TILE / 32);
+        const int lane_g = wid * 64 + (tid & 63);
+        for (int64_t it = 0; it < n_iter; it++) {
+            for (int task = lane_g; task < TASKS; task += NPROD * 64) {
+                const int g = task / (TILE / 32);
+                const int o = task % (TILE / 32);
+                if (b0 + g >= a.n) continue;
+                const int64_t off = it * TILE + (int64_t)o * 32;
+                uint32_t accp[P][8];
+#pragma unroll
+                for (int i = 0; i < P; i++)
+#pragma unroll
+                    for (int pb = 0; pb < 8; pb++) accp[i][pb] = 0;
+                uint32_t xc[8], xn[8];
+                {
+                    const uint8_t *row = a.data + ((b0 + g) * D) * stride + off;
+                    uint4 lo = *(const uint4 *)row;
+                    uint4 hi = *(const uint4 *)(row + 16);
+                    xc[0] = lo.x; xc[1] = lo.y; xc[2] = lo.z; xc[3] = lo.w;
+                    xc[4] = hi.x; xc[5] = hi.y; xc[6] = hi.z; xc[7] = hi.w;
+                }
+#pragma unroll
+                for (int k = 0; k < D; k++) {
+                    if (k + 1 < D) {
+                        const uint8_t *row =
+                            a.data + ((b0 + g) * D + k + 1) * stride + off;
+                        uint4 lo = *(const uint4 *)row;
+                        uint4 hi = *(const uint4 *)(row + 16);
+                        xn[0] = lo.x; xn[1] = lo.y; xn[2] = lo.z;
+                        xn[3] = lo.w; xn[4] = hi.x; xn[5] = hi.y;
+                        xn[6] = hi.z; xn[7] = hi.w;
+                    }
+                    bs_transpose(xc);
+                    bs_acc_k<D, P, MAT>(k, xc, accp);
+#pragma unroll
+                    for (int w = 0; w < 8; w++) xc[w] = xn[w];
+                }
+#pragma unroll
+                for (int i = 0; i < P; i++) {
+                    bs_transpose(accp[i]);
+                    uint8_t *orow =
+                        a.parity + ((b0 + g) * P + i) * stride + off;
+                    typedef unsigned int v4u
+                        __attribute__((ext_vector_type(4)));
+                    v4u vlo = {accp[i][0], accp[i][1], accp[i][2],
+                               accp[i][3]};
+                    v4u vhi = {accp[i][4], accp[i][5], accp[i][6],
+                               accp[i][7]};
+                    __builtin_nontemporal_store(vlo, (v4u *)orow);
+                    __builtin_nontemporal_store(vhi, (v4u *)(orow + 16));
+                }
+            }
+            if ((it + 1) % PUBK == 0 || it + 1 == n_iter) {
+                __builtin_amdgcn_s_waitcnt(0x0f70); /* vmcnt(0) */
+                if ((tid & 63) == 0)
+                    __hip_atomic_store(&prog[wid], (int)it + 1,
+                                       __ATOMIC_RELEASE,
+                                       __HIP_MEMORY_SCOPE_WORKGROUP);
+            }
+        }
+        return;
+    }
     if (wid < NPROD) {
         /* ---- producer: (block g, 16-B column o) tasks per tile ----
          * All of a task's D row loads are issued before any ladder math
@@ -468,6 +539,16 @@ extern "C" hipError_t mec_launch_fused3_encode_hh(int d, int p,
     if (d == D && p == P) {                                                  \
         constexpr int G8 = 4 < 64 / (D + P) ? 4 : 64 / (D + P);             \
         constexpr int G4 = 2 < 32 / (D + P) ? 2 : 32 / (D + P);             \
+        constexpr int G6 = (2 * 2) < (4 * 16 / (D + P))                     \
+                               ? (2 * 2) : (4 * 16 / (D + P));               \
+        if (G6 >= 1 && (f3cfg == 0 || f3cfg == 6) &&                         \
+            (args->n / G6 >= f3min || f3cfg == 6)) {                         \
+            dim3 grid((uint32_t)((args->n + G6 - 1) / G6));                  \
+            hipLaunchKernelGGL(                                              \
+                (fused3_encode_hh_kernel<D, P, MAT_##D##_##P, 6, 2, 1>),     \
+                grid, dim3(384), 0, stream, *args);                          \
+            return hipGetLastError();                                        \
+        }                                                                    \
         if (G8 >= 1 && f3cfg != 4 && (args->n / G8 >= f3min ||            \
                                        f3cfg == 8)) {                                \
             dim3 grid((uint32_t)((args->n + G8 - 1) / G8));                  \

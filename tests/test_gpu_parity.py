@@ -592,7 +592,8 @@ print("KNOB_OK")
 @pytest.mark.parametrize("knob", ["MEC_HH_LDS", "MEC_FUSED2", "MEC_FUSED",
                                   "MEC_HH_WG", "MEC_HH4_OFF", "MEC_SHA_WG",
                                   "MEC_FUSED3_OFF", "MEC_F3_MIN1",
-                                  "MEC_F3_MIN1_W4"])
+                                  "MEC_F3_MIN1_W4", "MEC_F3_MIN1_W6",
+                                  "MEC_GF_BS_OFF"])
 def test_knob_variants_bit_exact(knob):
     """The in-tree experiment knobs are env-latched at first
     use, so each variant runs in a subprocess.  Every knob'd kernel must
@@ -612,6 +613,11 @@ def test_knob_variants_bit_exact(knob):
     elif knob == "MEC_F3_MIN1_W4":
         env["MEC_F3_MIN"] = "1"  # force fused3 4-wave at ANY batch size
         env["MEC_F3_CFG"] = "4"
+    elif knob == "MEC_F3_MIN1_W6":
+        env["MEC_F3_MIN"] = "1"  # force fused3 6-wave (bit-sliced producer)
+        env["MEC_F3_CFG"] = "6"
+    elif knob == "MEC_GF_BS_OFF":
+        env["MEC_GF_BS"] = "0"   # xtime-ladder encode kernel
     else:
         env[knob] = "1"
     env["MEC_TEST_REPO"] = os.path.dirname(HERE)
