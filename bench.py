@@ -173,8 +173,12 @@ def main():
     # HighwayHash only: SHA-256's chains are so latency-bound that GF
     # co-residency slows them more than the overlap saves (13.4 vs 10.9
     # ms/step measured)
+    # r2: both legs run at their resource floors (GF memory-bound after
+    # bit-slicing, hash VALU-bound) — overlapping them mixes traffic at a
+    # worse joint rate than running them back-to-back (measured 0.657 vs
+    # 0.636 ms/step); MEC_PIPE=1 restores the r1 overlap for comparison
     use_pipe = (not is_decode) and algo in (2, 3) \
-        and os.environ.get("MEC_PIPE", "1") != "0"
+        and os.environ.get("MEC_PIPE", "0") == "1"
     step_no = [0]
 
     def step():

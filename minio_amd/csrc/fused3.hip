@@ -509,8 +509,14 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
 extern "C" hipError_t mec_launch_fused3_encode_hh(int d, int p,
                                                   const FusedArgs *args,
                                                   hipStream_t stream) {
+    /* Default OFF (r2): the workload is VALU-pipe-bound once the GF leg
+     * is bit-sliced — the kernel pair runs each leg at ~92% of its
+     * resource floor with 7-8 waves/SIMD of latency hiding, while the
+     * fused persistent layout caps at 1-2 waves/SIMD and pays 40-70%
+     * stall overhead (DESIGN.md r2 notes).  Kept in-tree, bit-exact and
+     * measured; enable with MEC_FUSED3=1. */
     static const char *env = getenv("MEC_FUSED3");
-    static const bool enabled = !env || atoi(env) != 0; /* default ON */
+    static const bool enabled = env && atoi(env) != 0;
     if (!enabled) return hipErrorNotSupported;
     if (args->shard_len % 1024 != 0) return hipErrorNotSupported;
     /* MEC_F3_CFG: force 8- or 4-wave config (perf sweeps); MEC_F3_MIN:

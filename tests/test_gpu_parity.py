@@ -590,6 +590,7 @@ print("KNOB_OK")
 
 
 @pytest.mark.parametrize("knob", ["MEC_HH_LDS", "MEC_FUSED2", "MEC_FUSED",
+                                  "MEC_FUSED3",
                                   "MEC_HH_WG", "MEC_HH4_OFF", "MEC_SHA_WG",
                                   "MEC_FUSED3_OFF", "MEC_F3_MIN1",
                                   "MEC_F3_MIN1_W4", "MEC_F3_MIN1_W6",
@@ -608,12 +609,15 @@ def test_knob_variants_bit_exact(knob):
     elif knob == "MEC_FUSED3_OFF":
         env["MEC_FUSED3"] = "0"  # two-kernel pair instead of fused v3
     elif knob == "MEC_F3_MIN1":
+        env["MEC_FUSED3"] = "1"
         env["MEC_F3_MIN"] = "1"  # force fused3 8-wave at ANY batch size
         env["MEC_F3_CFG"] = "8"
     elif knob == "MEC_F3_MIN1_W4":
+        env["MEC_FUSED3"] = "1"
         env["MEC_F3_MIN"] = "1"  # force fused3 4-wave at ANY batch size
         env["MEC_F3_CFG"] = "4"
     elif knob == "MEC_F3_MIN1_W6":
+        env["MEC_FUSED3"] = "1"
         env["MEC_F3_MIN"] = "1"  # force fused3 6-wave (bit-sliced producer)
         env["MEC_F3_CFG"] = "6"
     elif knob == "MEC_GF_BS_OFF":
