@@ -65,6 +65,22 @@ int hash_size(int algo) {
     }
 }
 
+/* 8x8 bit matrix of the GF(2^8)/0x11D linear map x -> c*x, packed as two
+ * dwords (byte b = rowmask: bit a set iff output bit b depends on input
+ * bit a) — the r2 bit-sliced matmul's runtime-matrix form (kernels.h). */
+void bs_pack_matrix(uint8_t c, uint32_t out[2]) {
+    uint64_t m = 0;
+    for (int b = 0; b < 8; b++) {
+        uint8_t row = 0;
+        for (int a = 0; a < 8; a++)
+            if ((mec::gf_mul(c, (uint8_t)(1u << a)) >> b) & 1)
+                row |= (uint8_t)(1u << a);
+        m |= (uint64_t)row << (8 * b);
+    }
+    out[0] = (uint32_t)m;
+    out[1] = (uint32_t)(m >> 32);
+}
+
 } // namespace
 
 struct mec_ctx {
@@ -85,7 +101,8 @@ struct mec_ctx {
     /* grow-only scratch (device + pinned host) for host-pointer calls */
     void *dev_a = nullptr, *dev_b = nullptr, *dev_c = nullptr;
     void *dev_d = nullptr; /* stream-assembly output */
-    size_t cap_a = 0, cap_b = 0, cap_c = 0, cap_d = 0;
+    void *dev_m = nullptr;  /* bit-matrix masks for the BS matmul (r2) */
+    size_t cap_a = 0, cap_b = 0, cap_c = 0, cap_d = 0, cap_m = 0;
     void *pin = nullptr;
     size_t cap_pin = 0;
 
@@ -269,6 +286,7 @@ void mec_ctx_destroy(mec_ctx *ctx) {
     if (ctx->dev_b) (void)hipFree(ctx->dev_b);
     if (ctx->dev_c) (void)hipFree(ctx->dev_c);
     if (ctx->dev_d) (void)hipFree(ctx->dev_d);
+    if (ctx->dev_m) (void)hipFree(ctx->dev_m);
     if (ctx->pin) (void)hipHostFree(ctx->pin);
     if (ctx->ev_start) (void)hipEventDestroy(ctx->ev_start);
     if (ctx->ev_stop) (void)hipEventDestroy(ctx->ev_stop);
@@ -598,6 +616,23 @@ static mec_status reconstruct_dev_locked(mec_ctx *ctx, int n,
         for (int i = 0; i < e; i++)
             for (int k = 0; k < d; k++)
                 a.mat[i * MEC_KMAX_D + k] = dec[(size_t)(t0 + i) * d + k];
+        /* bit-sliced path: pack the 8x8 bit matrices and upload (a few
+         * hundred bytes; synchronous — tiny next to the kernel) */
+        {
+            uint32_t masks[MEC_KMAX_E * MEC_KMAX_D * 2];
+            for (int i = 0; i < e; i++)
+                for (int k = 0; k < d; k++)
+                    bs_pack_matrix(a.mat[i * MEC_KMAX_D + k],
+                                   &masks[((size_t)i * d + k) * 2]);
+            size_t mb = (size_t)e * d * 2 * sizeof(uint32_t);
+            mec_status st2;
+            if ((st2 = ctx->ensure(&ctx->dev_m, &ctx->cap_m, mb)) != MEC_OK)
+                return st2;
+            HIP_TRY(hipStreamSynchronize(ctx->stream)); /* prior launch may
+                still read dev_m */
+            HIP_TRY(hipMemcpy(ctx->dev_m, masks, mb, hipMemcpyHostToDevice));
+            a.bs_masks = (const uint32_t *)ctx->dev_m;
+        }
         HIP_TRY(mec_launch_gf_matmul(&a, e, n, ctx->stream));
     }
     return MEC_OK;

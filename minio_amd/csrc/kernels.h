@@ -24,6 +24,11 @@ struct GfMatmulArgs {
     uint8_t src_rows[MEC_KMAX_D];
     uint8_t dst_rows[MEC_KMAX_E];
     uint8_t mat[MEC_KMAX_E * MEC_KMAX_D]; /* row t: coefficients over src */
+    /* bit-sliced path (r2): device pointer to E*d*2 u32 —
+     * bs_masks[(t*d+k)*2 + w] = dword w of the 8x8 bit matrix of
+     * mat[t][k] (byte b = rowmask(c,b): bit a set iff output bit b of
+     * c*x depends on input bit a).  NULL -> ladder kernel. */
+    const uint32_t *bs_masks;
 };
 
 /* mode: which shard rows this launch hashes.  Sums always land in the

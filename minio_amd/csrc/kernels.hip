@@ -305,6 +305,86 @@ __global__ void __launch_bounds__(256) gf_matmul_kernel(GfMatmulArgs a) {
     }
 }
 
+/* ---- bit-sliced generic matmul (reconstruct; runtime matrices) ---------
+ *
+ * Same plane-transpose trick as gf_encode_bs_kernel, but the decode
+ * matrix is only known at run time, so the per-(output,plane) XOR
+ * selection uses 0/~0 masks ((x & m) ^ acc as one v_bitop3) instead of
+ * compile-time folds: 64 VALU per (src,dst) pair per 32 B regardless of
+ * matrix density, vs the ladder's ~90-160.  Masks are built host-side
+ * (mec::bs_build_masks) and read through L1 (8 dwords per (t,k), hot
+ * after the first column). */
+template <int E>
+__global__ void __launch_bounds__(256) gf_matmul_bs_kernel(GfMatmulArgs a) {
+    const int b = blockIdx.y;
+    const int64_t cols = (a.shard_len + 31) / 32;
+    const uint8_t *__restrict__ sbase = a.src + (int64_t)b * a.src_item_stride;
+    uint8_t *__restrict__ obase = a.dst + (int64_t)b * a.dst_item_stride;
+
+    for (int64_t c = blockIdx.x * blockDim.x + threadIdx.x; c < cols;
+         c += (int64_t)gridDim.x * blockDim.x) {
+        const int64_t j = c * 32;
+        uint32_t accp[E][8];
+#pragma unroll
+        for (int i = 0; i < E; i++)
+#pragma unroll
+            for (int pb = 0; pb < 8; pb++) accp[i][pb] = 0;
+        uint32_t xc[8], xn[8];
+        {
+            const uint8_t *row = sbase + (int64_t)a.src_rows[0] * a.row_stride + j;
+            uint4 lo = *(const uint4 *)row;
+            uint4 hi = *(const uint4 *)(row + 16);
+            xc[0] = lo.x; xc[1] = lo.y; xc[2] = lo.z; xc[3] = lo.w;
+            xc[4] = hi.x; xc[5] = hi.y; xc[6] = hi.z; xc[7] = hi.w;
+        }
+        for (int k = 0; k < a.d; k++) {
+            if (k + 1 < a.d) {
+                const uint8_t *row =
+                    sbase + (int64_t)a.src_rows[k + 1] * a.row_stride + j;
+                uint4 lo = *(const uint4 *)row;
+                uint4 hi = *(const uint4 *)(row + 16);
+                xn[0] = lo.x; xn[1] = lo.y; xn[2] = lo.z; xn[3] = lo.w;
+                xn[4] = hi.x; xn[5] = hi.y; xn[6] = hi.z; xn[7] = hi.w;
+            }
+            bs_transpose(xc);
+#pragma unroll
+            for (int i = 0; i < E; i++) {
+                /* the 8x8 bit matrix of mat[i][k], 2 dwords; loaded once
+                 * per (i,k) and moved to SGPRs so the per-element mask
+                 * expansion runs on the SCALAR pipe concurrently with the
+                 * VALU — leaving exactly ONE v_bitop3 per matrix element */
+                const uint32_t *m = a.bs_masks + ((int64_t)i * a.d + k) * 2;
+                const uint32_t mlo = __builtin_amdgcn_readfirstlane(m[0]);
+                const uint32_t mhi = __builtin_amdgcn_readfirstlane(m[1]);
+#pragma unroll
+                for (int pb = 0; pb < 8; pb++) {
+                    uint32_t acc = accp[i][pb];
+                    const uint32_t mk =
+                        (pb < 4) ? (mlo >> (8 * pb)) : (mhi >> (8 * (pb - 4)));
+#pragma unroll
+                    for (int pa = 0; pa < 8; pa++) {
+                        const uint32_t mm = 0u - ((mk >> pa) & 1u);
+                        acc = (uint32_t)__builtin_amdgcn_bitop3_b32(
+                            xc[pa], mm, acc, 0x6a); /* (x & m) ^ acc */
+                    }
+                    accp[i][pb] = acc;
+                }
+            }
+#pragma unroll
+            for (int w = 0; w < 8; w++) xc[w] = xn[w];
+        }
+#pragma unroll
+        for (int i = 0; i < E; i++) {
+            bs_transpose(accp[i]);
+            uint8_t *orow = obase + (int64_t)a.dst_rows[i] * a.row_stride + j;
+            *(uint4 *)orow = uint4{accp[i][0], accp[i][1], accp[i][2],
+                                   accp[i][3]};
+            *(uint4 *)(orow + 16) = uint4{accp[i][4], accp[i][5],
+                                          accp[i][6], accp[i][7]};
+        }
+    }
+}
+
 /* ---- HighwayHash-256 (one chain per lane) ------------------------------
  * Portable algorithm as published (minio/highwayhash v1.0.3 semantics,
  * magic key passed from host; pinned by tests/golden/bitrot_selftest.json).
@@ -1347,6 +1427,28 @@ hipError_t mec_launch_gf_encode_spec(int d, int p, const GfEncArgs *args,
 
 hipError_t mec_launch_gf_matmul(const GfMatmulArgs *args, int n_dst, int n,
                                 hipStream_t stream) {
+    static const int env_bs = gf_env_int("MEC_GFM_BS", 1);
+    if (env_bs && args->bs_masks != nullptr) {
+        const int64_t cols32 = (args->shard_len + 31) / 32;
+        int64_t max_x = (cols32 + 255) / 256;
+        int64_t want_x = ((int64_t)2048 * 4 + n - 1) / n;
+        int64_t blocks_x = want_x < max_x ? want_x : max_x;
+        if (blocks_x < 1) blocks_x = 1;
+        dim3 grid((uint32_t)blocks_x, n);
+        dim3 blk(256);
+#define CASEB(E)                                                             \
+    case E:                                                                  \
+        hipLaunchKernelGGL((gf_matmul_bs_kernel<E>), grid, blk, 0, stream,   \
+                           *args);                                           \
+        return hipGetLastError();
+        switch (n_dst) {
+            CASEB(1) CASEB(2) CASEB(3) CASEB(4) CASEB(5) CASEB(6) CASEB(7)
+            CASEB(8)
+        default:
+            return hipErrorInvalidValue;
+        }
+#undef CASEB
+    }
     const int64_t cols = (args->shard_len + 15) >> 4;
     /* >=2048 workgroups total to fill 256 CUs, but never more than the work */
     int64_t max_x = (cols + 255) / 256;
