@@ -992,7 +992,11 @@ __global__ void __launch_bounds__(256) hh256_lds_kernel(HashArgs a) {
 
 /* ---- SHA-256 (one chain per lane), FIPS 180-4 -------------------------- */
 
-__constant__ uint32_t SHA_K[64] = {
+/* constexpr (not __constant__): the rounds are fully unrolled with
+ * literal indices, so K folds to inline literal operands — as a
+ * __constant__ buffer the compiler preloaded round constants into SGPRs
+ * (measured 44-88 spilled SGPRs in the NC=2 kernels, r2) */
+__device__ constexpr uint32_t SHA_K[64] = {
     0x428a2f98, 0x71374491, 0xb5c0fbcf, 0xe9b5dba5, 0x3956c25b, 0x59f111f1,
     0x923f82a4, 0xab1c5ed5, 0xd807aa98, 0x12835b01, 0x243185be, 0x550c7dc3,
     0x72be5d74, 0x80deb1fe, 0x9bdc06a7, 0xc19bf174, 0xe49b69c1, 0xefbe4786,
@@ -1300,47 +1304,61 @@ __global__ void __launch_bounds__(256) scatter_rows_kernel(ScatterArgs a) {
 
 __global__ void __launch_bounds__(256) stream_interleave_kernel(
     InterleaveArgs a) {
+    /* One workgroup per [hash||shard] entry (drive s, block b).  The
+     * entry's destination base (s*n+b)*pitch is byte-arbitrary whenever
+     * pitch = 32+S is odd (every ragged-S geometry, e.g. EC12+4), so the
+     * r1 version fell back to byte-wise copies for exactly those
+     * geometries.  Here stores are always aligned
+     * 16-B dwordx4 to the dst: a head/tail handles the odd edges
+     * byte-wise, and the body gathers each 16 output bytes from two
+     * aligned source loads sheared with v_alignbyte (constant shift per
+     * entry). */
     const int total = a.d + a.p;
-    const int64_t pitch = 32 + a.S;           /* per-block entry */
-    const int64_t units_per_entry = pitch / 16; /* S%16==0 on this path? S
-        is ceil(block/d) — may be ragged; guard below uses byte tail */
-    const int64_t nunits = (int64_t)total * a.n * ((pitch + 15) / 16);
-    const int64_t per_entry = (pitch + 15) / 16;
-    (void)units_per_entry;
-    for (int64_t u = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         u < nunits; u += (int64_t)gridDim.x * blockDim.x) {
-        const int s = (int)(u / (a.n * per_entry));
-        const int64_t r = u % (a.n * per_entry);
-        const int64_t b = r / per_entry;
-        const int64_t o = (r % per_entry) * 16;
-        uint8_t *dst = a.out + ((int64_t)s * a.n + b) * pitch + o;
-        if (o < 32) {
-            /* hash half: two 16-B units; dst is 16-B aligned only when
-             * pitch%16==0 (ragged S, e.g. EC12+4, makes entry bases odd) */
-            const uint8_t *h = a.sums + (b * total + s) * 32 + o;
-            if ((((int64_t)s * a.n + b) * pitch + o) % 16 == 0) {
-                *(uint4 *)dst = *(const uint4 *)h;
-            } else {
-                for (int i = 0; i < 16; i++) dst[i] = h[i];
-            }
+    const int64_t pitch = 32 + a.S;
+    const int64_t e = blockIdx.x;          /* entry: s = e/n, b = e%n */
+    const int s = (int)(e / a.n);
+    const int64_t b = e % a.n;
+    if (s >= total) return;
+    uint8_t *dst = a.out + e * pitch;
+    const uint8_t *h = a.sums + (b * total + s) * 32;
+    const uint8_t *row = (s < a.d)
+        ? a.data + (b * a.d + s) * a.row_stride
+        : a.parity + (b * a.p + (s - a.d)) * a.row_stride;
+    const int t = (int)threadIdx.x;
+    /* hash half: 32 B, byte-wise (arbitrary dst alignment, tiny) */
+    if (t < 32) dst[t] = h[t];
+    /* shard: dst offset 32, src row 16-B aligned */
+    uint8_t *sd = dst + 32;
+    const int head = (int)((16 - ((uintptr_t)sd & 15)) & 15);
+    const int64_t S = a.S;
+    /* head + tail bytes */
+    for (int64_t i = t; i < head && i < S; i += blockDim.x) sd[i] = row[i];
+    const int64_t units = (S - head) / 16;   /* full 16-B dst units */
+    for (int64_t i = head + units * 16 + t; i < S; i += blockDim.x)
+        sd[i] = row[i];
+    /* body: dst 16-B aligned; src at constant misalignment sh */
+    const int sh = head & 15;
+    const int dw = sh >> 2, bo = sh & 3;
+    for (int64_t u = t; u < units; u += blockDim.x) {
+        const int64_t off = head + u * 16;
+        const int64_t base = off & ~(int64_t)15;
+        if (base + 32 <= a.row_stride) {
+            uint4 q0 = *(const uint4 *)(row + base);
+            uint4 q1 = *(const uint4 *)(row + base + 16);
+            uint32_t d8[8] = {q0.x, q0.y, q0.z, q0.w, q1.x, q1.y, q1.z,
+                              q1.w};
+            uint4 o;
+            /* out dword i = bytes [off+4i, off+4i+4): alignbyte shears
+             * the concat {hi:lo} right by bo bytes */
+            o.x = __builtin_amdgcn_alignbyte(d8[dw + 1], d8[dw + 0], bo);
+            o.y = __builtin_amdgcn_alignbyte(d8[dw + 2], d8[dw + 1], bo);
+            o.z = __builtin_amdgcn_alignbyte(d8[dw + 3], d8[dw + 2], bo);
+            o.w = __builtin_amdgcn_alignbyte(d8[dw + 4], d8[dw + 3], bo);
+            *(uint4 *)(sd + off) = o;
         } else {
-            const int64_t j = o - 32;
-            const uint8_t *row = (s < a.d)
-                ? a.data + (b * a.d + s) * a.row_stride
-                : a.parity + (b * a.p + (s - a.d)) * a.row_stride;
-            if (j + 16 <= a.S) {
-                /* rows are 16-B aligned but dst is offset by 32 within a
-                 * pitch that may be odd-16 — dst IS 16-B aligned iff pitch
-                 * %16==0; handle bytewise when not */
-                if ((((int64_t)s * a.n + b) * pitch + o) % 16 == 0 &&
-                    (j % 16) == 0) {
-                    *(uint4 *)dst = *(const uint4 *)(row + j);
-                } else {
-                    for (int i = 0; i < 16; i++) dst[i] = row[j + i];
-                }
-            } else if (j < a.S) {
-                for (int i = 0; i < (int)(a.S - j); i++) dst[i] = row[j + i];
-            }
+            /* last unit of the last row may not have 32 readable source
+             * bytes within the stride: byte-wise */
+            for (int i2 = 0; i2 < 16; i2++) sd[off + i2] = row[off + i2];
         }
     }
 }
@@ -1490,12 +1508,8 @@ hipError_t mec_launch_scatter_rows(const ScatterArgs *args,
 
 hipError_t mec_launch_stream_interleave(const InterleaveArgs *args,
                                         hipStream_t stream) {
-    const int64_t pitch = 32 + args->S;
-    int64_t units = (int64_t)(args->d + args->p) * args->n *
-                    ((pitch + 15) / 16);
-    int64_t blocks = (units + 255) / 256;
-    if (blocks > 16384) blocks = 16384;
-    hipLaunchKernelGGL(stream_interleave_kernel, dim3((uint32_t)blocks),
+    int64_t entries = (int64_t)(args->d + args->p) * args->n;
+    hipLaunchKernelGGL(stream_interleave_kernel, dim3((uint32_t)entries),
                        dim3(256), 0, stream, *args);
     return hipGetLastError();
 }
