@@ -377,6 +377,20 @@ static mec_status encode_dev_locked(mec_ctx *ctx, int n, const void *data_dev,
             for (int k = 0; k < d; k++)
                 a.mat[i * MEC_KMAX_D + k] =
                     ctx->enc_matrix[(size_t)(d + i0 + i) * d + k];
+        {
+            uint32_t masks[MEC_KMAX_E * MEC_KMAX_D * 2];
+            for (int i = 0; i < e; i++)
+                for (int k = 0; k < d; k++)
+                    bs_pack_matrix(a.mat[i * MEC_KMAX_D + k],
+                                   &masks[((size_t)i * d + k) * 2]);
+            size_t mb = (size_t)e * d * 2 * sizeof(uint32_t);
+            mec_status st2;
+            if ((st2 = ctx->ensure(&ctx->dev_m, &ctx->cap_m, mb)) != MEC_OK)
+                return st2;
+            HIP_TRY(hipStreamSynchronize(ctx->stream));
+            HIP_TRY(hipMemcpy(ctx->dev_m, masks, mb, hipMemcpyHostToDevice));
+            a.bs_masks = (const uint32_t *)ctx->dev_m;
+        }
         HIP_TRY(mec_launch_gf_matmul(&a, e, n, ctx->stream));
     }
 gf_done:

@@ -1053,6 +1053,126 @@ __device__ void sha256_block(uint32_t h[8], const uint32_t w_in[16]) {
     h[4] += e; h[5] += f; h[6] += g; h[7] += hh;
 }
 
+/* 16-round-body variant: the fully-unrolled 64-round loop at NC=2 is a
+ * ~12 KiB body — big enough that co-resident waves thrash the 32 KiB
+ * I-cache (r2: NC2 at WG512 measured 1.8x SLOWER than 1 wave/SIMD).
+ * Rolling to a 16-round body (schedule indices stay static mod 16, the
+ * a..h renaming closes every 8 rounds) cuts the body 4x; K for rounds
+ * 16..63 comes from scalar loads of the rodata table. */
+template <int NC, int LB = 256>
+__global__ void __launch_bounds__(LB, 1) sha256_batch_r16_kernel(HashArgs a) {
+    const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t c0 = tid * NC;
+    if (c0 >= a.n_chains) return;
+
+    bool act[NC];
+    const uint8_t *mp[NC];
+    int64_t sum_idx[NC];
+    uint32_t h[NC][8];
+#pragma unroll
+    for (int u = 0; u < NC; u++) {
+        act[u] = c0 + u < a.n_chains;
+        mp[u] = chain_ptr(a, act[u] ? c0 + u : c0, sum_idx[u]);
+        const uint32_t iv[8] = {0x6a09e667, 0xbb67ae85, 0x3c6ef372,
+                                0xa54ff53a, 0x510e527f, 0x9b05688c,
+                                0x1f83d9ab, 0x5be0cd19};
+#pragma unroll
+        for (int i = 0; i < 8; i++) h[u][i] = iv[i];
+    }
+    int64_t len = a.msg_len;
+    while (len >= 64) {
+        uint32_t w[NC][16];
+#pragma unroll
+        for (int u = 0; u < NC; u++) {
+            const uint4 *p = (const uint4 *)mp[u];
+#pragma unroll
+            for (int q = 0; q < 4; q++) {
+                uint4 v = p[q];
+                w[u][4 * q + 0] = bswap32(v.x);
+                w[u][4 * q + 1] = bswap32(v.y);
+                w[u][4 * q + 2] = bswap32(v.z);
+                w[u][4 * q + 3] = bswap32(v.w);
+            }
+            mp[u] += 64;
+        }
+        uint32_t A[NC], B[NC], C[NC], D[NC], E[NC], F[NC], G[NC], H[NC];
+#pragma unroll
+        for (int u = 0; u < NC; u++) {
+            A[u] = h[u][0]; B[u] = h[u][1]; C[u] = h[u][2]; D[u] = h[u][3];
+            E[u] = h[u][4]; F[u] = h[u][5]; G[u] = h[u][6]; H[u] = h[u][7];
+        }
+#define SHA_ROUND(u, wi, kk)                                                 \
+        {                                                                    \
+            uint32_t S1 = rotr32(E[u], 6) ^ rotr32(E[u], 11) ^               \
+                          rotr32(E[u], 25);                                  \
+            uint32_t ch = (E[u] & F[u]) ^ (~E[u] & G[u]);                    \
+            uint32_t t1 = H[u] + S1 + ch + (kk) + (wi);                      \
+            uint32_t S0 = rotr32(A[u], 2) ^ rotr32(A[u], 13) ^               \
+                          rotr32(A[u], 22);                                  \
+            uint32_t maj = (A[u] & B[u]) ^ (A[u] & C[u]) ^ (B[u] & C[u]);    \
+            uint32_t t2 = S0 + maj;                                          \
+            H[u] = G[u]; G[u] = F[u]; F[u] = E[u]; E[u] = D[u] + t1;         \
+            D[u] = C[u]; C[u] = B[u]; B[u] = A[u]; A[u] = t1 + t2;           \
+        }
+        /* rounds 0..15: message words direct, K folds to literals */
+#pragma unroll
+        for (int r = 0; r < 16; r++)
+#pragma unroll
+            for (int u = 0; u < NC; u++) SHA_ROUND(u, w[u][r], SHA_K[r]);
+        /* rounds 16..63: three passes of a 16-round body */
+#pragma unroll 1
+        for (int q = 1; q < 4; q++) {
+#pragma unroll
+            for (int r = 0; r < 16; r++) {
+                const uint32_t kk = SHA_K[(q << 4) + r];
+#pragma unroll
+                for (int u = 0; u < NC; u++) {
+                    uint32_t w15 = w[u][(r + 1) & 15], w2 = w[u][(r + 14) & 15];
+                    uint32_t s0 = rotr32(w15, 7) ^ rotr32(w15, 18) ^ (w15 >> 3);
+                    uint32_t s1 = rotr32(w2, 17) ^ rotr32(w2, 19) ^ (w2 >> 10);
+                    uint32_t wi = w[u][r] + s0 + w[u][(r + 9) & 15] + s1;
+                    w[u][r] = wi;
+                    SHA_ROUND(u, wi, kk);
+                }
+            }
+        }
+#undef SHA_ROUND
+#pragma unroll
+        for (int u = 0; u < NC; u++) {
+            h[u][0] += A[u]; h[u][1] += B[u]; h[u][2] += C[u]; h[u][3] += D[u];
+            h[u][4] += E[u]; h[u][5] += F[u]; h[u][6] += G[u]; h[u][7] += H[u];
+        }
+        len -= 64;
+    }
+    /* tail: same as sha256_batch_kernel */
+#pragma unroll
+    for (int u = 0; u < NC; u++) {
+        uint8_t tail[128];
+#pragma unroll
+        for (int i = 0; i < 128; i++) tail[i] = 0;
+        for (int i = 0; i < (int)len; i++) tail[i] = mp[u][i];
+        tail[(int)len] = 0x80;
+        const int tlen = (len < 56) ? 64 : 128;
+        uint64_t bits = (uint64_t)a.msg_len * 8;
+        for (int i = 0; i < 8; i++)
+            tail[tlen - 1 - i] = (uint8_t)(bits >> (8 * i));
+        for (int blk = 0; blk < tlen; blk += 64) {
+            uint32_t wt[16];
+            for (int i = 0; i < 16; i++) {
+                const uint8_t *q = tail + blk + 4 * i;
+                wt[i] = ((uint32_t)q[0] << 24) | ((uint32_t)q[1] << 16) |
+                        ((uint32_t)q[2] << 8) | q[3];
+            }
+            sha256_block(h[u], wt);
+        }
+        if (act[u]) {
+            uint8_t *out = a.sums + sum_idx[u] * 32;
+            for (int i = 0; i < 8; i++)
+                *(uint32_t *)(out + 4 * i) = bswap32(h[u][i]);
+        }
+    }
+}
+
 /* NC independent chains per lane: SHA-256's round chain is strictly serial
  * (measured ~17 cyc/instr effective at 1 chain/lane, 1 wave/SIMD — pure
  * dependency latency); interleaving NC chains fills the stalls. */
@@ -1538,6 +1658,21 @@ hipError_t mec_launch_hash(int algo, const HashArgs *args,
              * for co-resident waves that fill each other's stalls is the
              * remaining occupancy lever (r2 probe). */
             static const int swg = gf_env_int("MEC_SHA_WG", 256);
+            static const int r16 = gf_env_int("MEC_SHA_R16", 0);
+            if (r16) {
+                if (swg >= 512 && nc == 2) {
+                    dim3 b512(512);
+                    grid.x =
+                        (uint32_t)(((args->n_chains + 1) / 2 + 511) / 512);
+                    hipLaunchKernelGGL((sha256_batch_r16_kernel<2, 512>),
+                                       grid, b512, 0, stream, *args);
+                    break;
+                }
+                grid.x = (uint32_t)(((args->n_chains + 1) / 2 + 255) / 256);
+                hipLaunchKernelGGL((sha256_batch_r16_kernel<2>), grid, blk,
+                                   0, stream, *args);
+                break;
+            }
             if (swg >= 512 && nc == 2) {
                 dim3 b512(512);
                 grid.x = (uint32_t)(((args->n_chains + 1) / 2 + 511) / 512);

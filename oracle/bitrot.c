@@ -198,7 +198,7 @@ double mo_cpu_reconstruct_bench(int d, int p, size_t block_len, int n_blocks,
         uint8_t *shards[MO_MAX_SHARDS];
         for (int s = 0; s < total; s++)
             shards[s] = bufs + ((size_t)b * total + s) * shard_len;
-        mo_rs_reconstruct(&rs, shards, present, shard_len, 1);
+        mo_rs_reconstruct_fast(&rs, shards, present, shard_len, 1);
     }
     double el = now_sec() - t0;
     volatile uint8_t sink = bufs[0];

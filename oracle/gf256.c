@@ -201,3 +201,47 @@ int mo_rs_reconstruct(const mo_rs *rs, uint8_t *const *shards,
     }
     return 0;
 }
+
+/* BENCH LEG ONLY (simd.c dispatch): reconstruct with the SIMD
+ * constant-multiply.  The checker path above stays scalar; bit-equality
+ * of the two is pinned by test_simd_matches_scalar (the GF op) plus the
+ * round-trip identity. */
+int mo_rs_reconstruct_fast(const mo_rs *rs, uint8_t *const *shards,
+                           const uint8_t *present, size_t shard_len,
+                           int data_only) {
+    int d = rs->d, total = rs->d + rs->p;
+    int n_present = 0;
+    for (int i = 0; i < total; i++)
+        if (present[i]) n_present++;
+    if (n_present == total) return 0;
+    if (n_present < d) return -2;
+    static _Thread_local gmat sub, dec;
+    int src_idx[MO_MAX_SHARDS];
+    sub.rows = sub.cols = d;
+    int r = 0;
+    for (int i = 0; i < total && r < d; i++) {
+        if (!present[i]) continue;
+        memcpy(sub.m[r], rs->matrix + (size_t)i * d, (size_t)d);
+        src_idx[r] = i;
+        r++;
+    }
+    if (gm_invert(&sub, &dec) != 0) return -1;
+    for (int t = 0; t < d; t++) {
+        if (present[t]) continue;
+        uint8_t *out = shards[t];
+        memset(out, 0, shard_len);
+        for (int k = 0; k < d; k++)
+            mo_gal_mul_xor_fast(dec.m[t][k], shards[src_idx[k]], out,
+                                shard_len);
+    }
+    if (data_only) return 0;
+    for (int t = d; t < total; t++) {
+        if (present[t]) continue;
+        uint8_t *out = shards[t];
+        memset(out, 0, shard_len);
+        const uint8_t *row = rs->matrix + (size_t)t * d;
+        for (int k = 0; k < d; k++)
+            mo_gal_mul_xor_fast(row[k], shards[k], out, shard_len);
+    }
+    return 0;
+}

@@ -103,6 +103,9 @@ void mo_gal_mul_xor_fast(uint8_t c, const uint8_t *in, uint8_t *out,
                          size_t n);
 void mo_rs_encode_fast(const mo_rs *rs, uint8_t *const *shards,
                        size_t shard_len);
+int mo_rs_reconstruct_fast(const mo_rs *rs, uint8_t *const *shards,
+                           const uint8_t *present, size_t shard_len,
+                           int data_only);
 void mo_hh256_fast(const uint8_t key[32], const uint8_t *msg, size_t len,
                    uint8_t out[32]);
 void mo_bitrot_sum_fast(int algo, const uint8_t *msg, size_t len,

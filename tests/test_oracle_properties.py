@@ -127,3 +127,37 @@ def test_simd_matches_scalar():
         msg = oracle.fill_random(ln, 0xABCDEF + ln)
         assert oracle.hh256_fast(key, msg) == oracle.bitrot_sum(
             oracle.HIGHWAYHASH256S, msg), f"len {ln} (isa={isa})"
+
+
+def test_simd_reconstruct_matches_scalar():
+    """mo_rs_reconstruct_fast (bench leg) == scalar checker on random
+    erasure patterns."""
+    import ctypes
+    import random
+    lib = oracle._lib
+    lib.mo_rs_reconstruct_fast.argtypes = [
+        ctypes.POINTER(oracle._MoRS), ctypes.POINTER(ctypes.POINTER(ctypes.c_uint8)),
+        ctypes.c_char_p, ctypes.c_size_t, ctypes.c_int]
+    lib.mo_rs_reconstruct_fast.restype = ctypes.c_int
+    rng = random.Random(5)
+    for (d, p) in [(8, 4), (12, 4), (4, 2)]:
+        bs = d * 997
+        data = oracle.fill_random(bs, d * 7 + p)
+        rs = oracle.RS(d, p)
+        want = rs.encode_data(data)
+        S = len(want[0])
+        for _ in range(6):
+            n_er = rng.randint(1, p)
+            erased = rng.sample(range(d + p), n_er)
+            bufs = [ctypes.create_string_buffer(
+                        b"" if i in erased else want[i], S)
+                    for i in range(d + p)]
+            arr = (ctypes.POINTER(ctypes.c_uint8) * (d + p))(
+                *[ctypes.cast(b, ctypes.POINTER(ctypes.c_uint8))
+                  for b in bufs])
+            present = bytes(0 if i in erased else 1 for i in range(d + p))
+            rc = lib.mo_rs_reconstruct_fast(ctypes.byref(rs._rs), arr,
+                                            present, S, 0)
+            assert rc == 0
+            for i in range(d + p):
+                assert bufs[i].raw[:S] == want[i], (d, p, erased, i)
