@@ -1080,20 +1080,38 @@ __global__ void __launch_bounds__(LB, 1) sha256_batch_r16_kernel(HashArgs a) {
         for (int i = 0; i < 8; i++) h[u][i] = iv[i];
     }
     int64_t len = a.msg_len;
-    while (len >= 64) {
-        uint32_t w[NC][16];
+    /* message double-buffer: the next 64-B block's loads ride under this
+     * block's 64 rounds instead of stalling at each block boundary */
+    uint4 mb[NC][4], mbn[NC][4];
+    if (len >= 64) {
 #pragma unroll
         for (int u = 0; u < NC; u++) {
             const uint4 *p = (const uint4 *)mp[u];
 #pragma unroll
-            for (int q = 0; q < 4; q++) {
-                uint4 v = p[q];
-                w[u][4 * q + 0] = bswap32(v.x);
-                w[u][4 * q + 1] = bswap32(v.y);
-                w[u][4 * q + 2] = bswap32(v.z);
-                w[u][4 * q + 3] = bswap32(v.w);
-            }
+            for (int q = 0; q < 4; q++) mb[u][q] = p[q];
             mp[u] += 64;
+        }
+    }
+    while (len >= 64) {
+        if (len >= 128) {
+#pragma unroll
+            for (int u = 0; u < NC; u++) {
+                const uint4 *p = (const uint4 *)mp[u];
+#pragma unroll
+                for (int q = 0; q < 4; q++) mbn[u][q] = p[q];
+                mp[u] += 64;
+            }
+        }
+        uint32_t w[NC][16];
+#pragma unroll
+        for (int u = 0; u < NC; u++) {
+#pragma unroll
+            for (int q = 0; q < 4; q++) {
+                w[u][4 * q + 0] = bswap32(mb[u][q].x);
+                w[u][4 * q + 1] = bswap32(mb[u][q].y);
+                w[u][4 * q + 2] = bswap32(mb[u][q].z);
+                w[u][4 * q + 3] = bswap32(mb[u][q].w);
+            }
         }
         uint32_t A[NC], B[NC], C[NC], D[NC], E[NC], F[NC], G[NC], H[NC];
 #pragma unroll
@@ -1103,13 +1121,15 @@ __global__ void __launch_bounds__(LB, 1) sha256_batch_r16_kernel(HashArgs a) {
         }
 #define SHA_ROUND(u, wi, kk)                                                 \
         {                                                                    \
-            uint32_t S1 = rotr32(E[u], 6) ^ rotr32(E[u], 11) ^               \
-                          rotr32(E[u], 25);                                  \
-            uint32_t ch = (E[u] & F[u]) ^ (~E[u] & G[u]);                    \
+            uint32_t S1 = xor3(rotr32(E[u], 6), rotr32(E[u], 11),            \
+                               rotr32(E[u], 25));                            \
+            uint32_t ch = (uint32_t)__builtin_amdgcn_bitop3_b32(             \
+                E[u], F[u], G[u], 0xca);                                     \
             uint32_t t1 = H[u] + S1 + ch + (kk) + (wi);                      \
-            uint32_t S0 = rotr32(A[u], 2) ^ rotr32(A[u], 13) ^               \
-                          rotr32(A[u], 22);                                  \
-            uint32_t maj = (A[u] & B[u]) ^ (A[u] & C[u]) ^ (B[u] & C[u]);    \
+            uint32_t S0 = xor3(rotr32(A[u], 2), rotr32(A[u], 13),            \
+                               rotr32(A[u], 22));                            \
+            uint32_t maj = (uint32_t)__builtin_amdgcn_bitop3_b32(            \
+                A[u], B[u], C[u], 0xe8);                                     \
             uint32_t t2 = S0 + maj;                                          \
             H[u] = G[u]; G[u] = F[u]; F[u] = E[u]; E[u] = D[u] + t1;         \
             D[u] = C[u]; C[u] = B[u]; B[u] = A[u]; A[u] = t1 + t2;           \
@@ -1128,8 +1148,10 @@ __global__ void __launch_bounds__(LB, 1) sha256_batch_r16_kernel(HashArgs a) {
 #pragma unroll
                 for (int u = 0; u < NC; u++) {
                     uint32_t w15 = w[u][(r + 1) & 15], w2 = w[u][(r + 14) & 15];
-                    uint32_t s0 = rotr32(w15, 7) ^ rotr32(w15, 18) ^ (w15 >> 3);
-                    uint32_t s1 = rotr32(w2, 17) ^ rotr32(w2, 19) ^ (w2 >> 10);
+                    uint32_t s0 = xor3(rotr32(w15, 7), rotr32(w15, 18),
+                                       w15 >> 3);
+                    uint32_t s1 = xor3(rotr32(w2, 17), rotr32(w2, 19),
+                                       w2 >> 10);
                     uint32_t wi = w[u][r] + s0 + w[u][(r + 9) & 15] + s1;
                     w[u][r] = wi;
                     SHA_ROUND(u, wi, kk);
@@ -1142,8 +1164,19 @@ __global__ void __launch_bounds__(LB, 1) sha256_batch_r16_kernel(HashArgs a) {
             h[u][0] += A[u]; h[u][1] += B[u]; h[u][2] += C[u]; h[u][3] += D[u];
             h[u][4] += E[u]; h[u][5] += F[u]; h[u][6] += G[u]; h[u][7] += H[u];
         }
+        if (len >= 128) {
+#pragma unroll
+            for (int u = 0; u < NC; u++)
+#pragma unroll
+                for (int q = 0; q < 4; q++) mb[u][q] = mbn[u][q];
+        }
         len -= 64;
     }
+    /* the double-buffer advanced mp one block past the remainder */
+#pragma unroll
+    for (int u = 0; u < NC; u++)
+        mp[u] = chain_ptr(a, act[u] ? c0 + u : c0, sum_idx[u]) +
+                (a.msg_len - len);
     /* tail: same as sha256_batch_kernel */
 #pragma unroll
     for (int u = 0; u < NC; u++) {
@@ -1197,20 +1230,38 @@ __global__ void __launch_bounds__(LB, 1) sha256_batch_kernel(HashArgs a) {
         for (int i = 0; i < 8; i++) h[u][i] = iv[i];
     }
     int64_t len = a.msg_len;
-    while (len >= 64) {
-        uint32_t w[NC][16];
+    /* message double-buffer: the next 64-B block's loads ride under this
+     * block's 64 rounds instead of stalling at each block boundary */
+    uint4 mb[NC][4], mbn[NC][4];
+    if (len >= 64) {
 #pragma unroll
         for (int u = 0; u < NC; u++) {
             const uint4 *p = (const uint4 *)mp[u];
 #pragma unroll
-            for (int q = 0; q < 4; q++) {
-                uint4 v = p[q];
-                w[u][4 * q + 0] = bswap32(v.x);
-                w[u][4 * q + 1] = bswap32(v.y);
-                w[u][4 * q + 2] = bswap32(v.z);
-                w[u][4 * q + 3] = bswap32(v.w);
-            }
+            for (int q = 0; q < 4; q++) mb[u][q] = p[q];
             mp[u] += 64;
+        }
+    }
+    while (len >= 64) {
+        if (len >= 128) {
+#pragma unroll
+            for (int u = 0; u < NC; u++) {
+                const uint4 *p = (const uint4 *)mp[u];
+#pragma unroll
+                for (int q = 0; q < 4; q++) mbn[u][q] = p[q];
+                mp[u] += 64;
+            }
+        }
+        uint32_t w[NC][16];
+#pragma unroll
+        for (int u = 0; u < NC; u++) {
+#pragma unroll
+            for (int q = 0; q < 4; q++) {
+                w[u][4 * q + 0] = bswap32(mb[u][q].x);
+                w[u][4 * q + 1] = bswap32(mb[u][q].y);
+                w[u][4 * q + 2] = bswap32(mb[u][q].z);
+                w[u][4 * q + 3] = bswap32(mb[u][q].w);
+            }
         }
         uint32_t A[NC], B[NC], C[NC], D[NC], E[NC], F[NC], G[NC], H[NC];
 #pragma unroll
