@@ -1298,6 +1298,12 @@ __global__ void __launch_bounds__(LB, 1) sha256_batch_kernel(HashArgs a) {
             h[u][0] += A[u]; h[u][1] += B[u]; h[u][2] += C[u]; h[u][3] += D[u];
             h[u][4] += E[u]; h[u][5] += F[u]; h[u][6] += G[u]; h[u][7] += H[u];
         }
+        if (len >= 128) {
+#pragma unroll
+            for (int u = 0; u < NC; u++)
+#pragma unroll
+                for (int q = 0; q < 4; q++) mb[u][q] = mbn[u][q];
+        }
         len -= 64;
     }
     /* tail: rem bytes + 0x80 pad + 8-byte big-endian bit length */
@@ -1709,7 +1715,7 @@ hipError_t mec_launch_hash(int algo, const HashArgs *args,
              * for co-resident waves that fill each other's stalls is the
              * remaining occupancy lever (r2 probe). */
             static const int swg = gf_env_int("MEC_SHA_WG", 256);
-            static const int r16 = gf_env_int("MEC_SHA_R16", 0);
+            static const int r16 = gf_env_int("MEC_SHA_R16", 1);
             if (r16) {
                 if (swg >= 512 && nc == 2) {
                     dim3 b512(512);
