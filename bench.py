@@ -259,6 +259,9 @@ def main():
             # kernels run sequentially on one stream, so the hash leg is
             # the fused-minus-gf difference
             t_hash = max(t_fused - t_gf, 1e-9)
+            # r2 note: the GF leg is HBM-bound (SQ: 10% active-issue, 90%
+            # waits at 7 waves/SIMD) while the hash leg is VALU-PIPE-bound
+            # (SQ: 88% active-issue) — see profiles/r2_sq.txt
             # per-launch algorithmic bytes:
             gf_bytes = n * (bs + p * S)          # read data, write parity
             hash_bytes = n * (total * S)         # read every shard once
@@ -278,7 +281,7 @@ def main():
             "achieved": round(achieved / 1e9, 1), "peak": peak / 1e9,
             "unit": "GB/s", "frac": round(achieved / peak, 4),
             "traffic": traffic,
-            "traffic_evidence": "profiles/r06_pmc_*.txt" if traffic else None,
+            "traffic_evidence": "profiles/r2_fetch.txt,r2_write.txt" if traffic else None,
         }
         if not is_decode:
             roofline["legs_ms"] = {"gf": round(t_gf * 1e3, 3),
