@@ -185,7 +185,9 @@ __global__ void __launch_bounds__(256) gf_encode_kernel(GfEncArgs a) {
 
 #include "gf_bs.h"
 
-template <int D, int P, const uint8_t (&MAT)[P][D], bool NT>
+/* PF: rows of load prefetch ahead of the transpose+fold (1 = r2 default;
+ * 2 = deeper cover now that bit-slicing cut the compute between loads) */
+template <int D, int P, const uint8_t (&MAT)[P][D], bool NT, int PF = 1>
 __global__ void __launch_bounds__(256) gf_encode_bs_kernel(GfEncArgs a) {
     const int b = blockIdx.y;
     const int64_t cols = (a.shard_len + 31) / 32;
@@ -200,29 +202,35 @@ __global__ void __launch_bounds__(256) gf_encode_bs_kernel(GfEncArgs a) {
         for (int i = 0; i < P; i++)
 #pragma unroll
             for (int pb = 0; pb < 8; pb++) accp[i][pb] = 0;
-        /* software pipeline: row k+1's loads in flight during row k's
-         * transpose + plane xors */
-        uint32_t xc[8], xn[8];
-        {
-            const uint8_t *row = sbase + j;
+        /* software pipeline: the next PF rows' loads in flight during
+         * this row's transpose + plane xors */
+        uint32_t xq[PF + 1][8];
+#pragma unroll
+        for (int pf = 0; pf < PF && pf < D; pf++) {
+            const uint8_t *row = sbase + (int64_t)pf * a.row_stride + j;
             uint4 lo = *(const uint4 *)row;
             uint4 hi = *(const uint4 *)(row + 16);
-            xc[0] = lo.x; xc[1] = lo.y; xc[2] = lo.z; xc[3] = lo.w;
-            xc[4] = hi.x; xc[5] = hi.y; xc[6] = hi.z; xc[7] = hi.w;
+            xq[pf][0] = lo.x; xq[pf][1] = lo.y;
+            xq[pf][2] = lo.z; xq[pf][3] = lo.w;
+            xq[pf][4] = hi.x; xq[pf][5] = hi.y;
+            xq[pf][6] = hi.z; xq[pf][7] = hi.w;
         }
 #pragma unroll
         for (int k = 0; k < D; k++) {
-            if (k + 1 < D) {
-                const uint8_t *row = sbase + (int64_t)(k + 1) * a.row_stride + j;
+            const int cur = k % (PF + 1);
+            if (k + PF < D) {
+                const int nxt = (k + PF) % (PF + 1);
+                const uint8_t *row =
+                    sbase + (int64_t)(k + PF) * a.row_stride + j;
                 uint4 lo = *(const uint4 *)row;
                 uint4 hi = *(const uint4 *)(row + 16);
-                xn[0] = lo.x; xn[1] = lo.y; xn[2] = lo.z; xn[3] = lo.w;
-                xn[4] = hi.x; xn[5] = hi.y; xn[6] = hi.z; xn[7] = hi.w;
+                xq[nxt][0] = lo.x; xq[nxt][1] = lo.y;
+                xq[nxt][2] = lo.z; xq[nxt][3] = lo.w;
+                xq[nxt][4] = hi.x; xq[nxt][5] = hi.y;
+                xq[nxt][6] = hi.z; xq[nxt][7] = hi.w;
             }
-            bs_transpose(xc);
-            bs_acc_k<D, P, MAT>(k, xc, accp);
-#pragma unroll
-            for (int w = 0; w < 8; w++) xc[w] = xn[w];
+            bs_transpose(xq[cur]);
+            bs_acc_k<D, P, MAT>(k, xq[cur], accp);
         }
 #pragma unroll
         for (int i = 0; i < P; i++) {
@@ -1558,17 +1566,28 @@ hipError_t mec_launch_gf_encode_spec(int d, int p, const GfEncArgs *args,
         /* bit-sliced encode (r2 default): ~2.4x fewer VALU slots than the
          * xtime ladder; 32 B per lane */
         const int64_t cols = (args->shard_len + 31) / 32;
+        static const int bswgx = gf_env_int("MEC_GF_BSWGX", 4);
         int64_t max_x = (cols + 255) / 256;
-        int64_t want_x = ((int64_t)2048 * 4 + n - 1) / n;
+        int64_t want_x = ((int64_t)2048 * bswgx + n - 1) / n;
         int64_t blocks_x = want_x < max_x ? want_x : max_x;
         if (blocks_x < 1) blocks_x = 1;
         dim3 grid((uint32_t)blocks_x, n);
         dim3 blk(256);
+        static const int bspf = gf_env_int("MEC_GF_BSPF", 1);
 #define XBS(D, P)                                                            \
         if (d == D && p == P) {                                              \
-            hipLaunchKernelGGL((gf_encode_bs_kernel<D, P, MAT_##D##_##P,     \
-                                                    true>),                  \
-                               grid, blk, 0, stream, *args);                 \
+            if (bspf >= 4)                                                   \
+                hipLaunchKernelGGL(                                          \
+                    (gf_encode_bs_kernel<D, P, MAT_##D##_##P, true, 4>),     \
+                    grid, blk, 0, stream, *args);                            \
+            else if (bspf == 2)                                              \
+                hipLaunchKernelGGL(                                          \
+                    (gf_encode_bs_kernel<D, P, MAT_##D##_##P, true, 2>),     \
+                    grid, blk, 0, stream, *args);                            \
+            else                                                             \
+                hipLaunchKernelGGL(                                          \
+                    (gf_encode_bs_kernel<D, P, MAT_##D##_##P, true, 1>),     \
+                    grid, blk, 0, stream, *args);                            \
             return hipGetLastError();                                        \
         }
         MEC_SPECIALIZED_GEOS(XBS)
