@@ -1566,14 +1566,14 @@ hipError_t mec_launch_gf_encode_spec(int d, int p, const GfEncArgs *args,
         /* bit-sliced encode (r2 default): ~2.4x fewer VALU slots than the
          * xtime ladder; 32 B per lane */
         const int64_t cols = (args->shard_len + 31) / 32;
-        static const int bswgx = gf_env_int("MEC_GF_BSWGX", 4);
+        static const int bswgx = gf_env_int("MEC_GF_BSWGX", 8);
         int64_t max_x = (cols + 255) / 256;
         int64_t want_x = ((int64_t)2048 * bswgx + n - 1) / n;
         int64_t blocks_x = want_x < max_x ? want_x : max_x;
         if (blocks_x < 1) blocks_x = 1;
         dim3 grid((uint32_t)blocks_x, n);
         dim3 blk(256);
-        static const int bspf = gf_env_int("MEC_GF_BSPF", 1);
+        static const int bspf = gf_env_int("MEC_GF_BSPF", 2);
 #define XBS(D, P)                                                            \
         if (d == D && p == P) {                                              \
             if (bspf >= 4)                                                   \
