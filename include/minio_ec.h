@@ -56,8 +56,6 @@ typedef enum {
     MEC_ERR_INVALID_ARG = 6,
     MEC_ERR_HIP = 7,            /* HIP runtime failure (message via
                                    mec_last_error) */
-    MEC_ERR_INTERNAL = 9,       /* invariant violation inside this library
-                                   (a bug, not a caller/quorum condition) */
     MEC_ERR_NO_GPU = 8,         /* no MI355X visible — the product path
                                    fails loudly, it never falls back */
     MEC_ERR_INTERNAL = 9,       /* invariant violation inside this library
