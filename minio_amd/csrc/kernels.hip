@@ -322,7 +322,7 @@ __global__ void __launch_bounds__(256) gf_matmul_kernel(GfMatmulArgs a) {
  * matrix density, vs the ladder's ~90-160.  Masks are built host-side
  * (mec::bs_build_masks) and read through L1 (8 dwords per (t,k), hot
  * after the first column). */
-template <int E>
+template <int E, int BRANCHY = 0>
 __global__ void __launch_bounds__(256) gf_matmul_bs_kernel(GfMatmulArgs a) {
     const int b = blockIdx.y;
     const int64_t cols = (a.shard_len + 31) / 32;
@@ -360,7 +360,10 @@ __global__ void __launch_bounds__(256) gf_matmul_bs_kernel(GfMatmulArgs a) {
                 /* the 8x8 bit matrix of mat[i][k], 2 dwords; loaded once
                  * per (i,k) and moved to SGPRs so the per-element mask
                  * expansion runs on the SCALAR pipe concurrently with the
-                 * VALU — leaving exactly ONE v_bitop3 per matrix element */
+                 * VALU — leaving exactly ONE v_bitop3 per matrix element.
+                 * BRANCHY=1 instead skips zero bits with wave-uniform
+                 * scalar branches (~half the VALU at ~density 0.5, paid
+                 * in s_cbranch overhead — measured) */
                 const uint32_t *m = a.bs_masks + ((int64_t)i * a.d + k) * 2;
                 const uint32_t mlo = __builtin_amdgcn_readfirstlane(m[0]);
                 const uint32_t mhi = __builtin_amdgcn_readfirstlane(m[1]);
@@ -369,11 +372,17 @@ __global__ void __launch_bounds__(256) gf_matmul_bs_kernel(GfMatmulArgs a) {
                     uint32_t acc = accp[i][pb];
                     const uint32_t mk =
                         (pb < 4) ? (mlo >> (8 * pb)) : (mhi >> (8 * (pb - 4)));
+                    if (BRANCHY) {
 #pragma unroll
-                    for (int pa = 0; pa < 8; pa++) {
-                        const uint32_t mm = 0u - ((mk >> pa) & 1u);
-                        acc = (uint32_t)__builtin_amdgcn_bitop3_b32(
-                            xc[pa], mm, acc, 0x6a); /* (x & m) ^ acc */
+                        for (int pa = 0; pa < 8; pa++)
+                            if ((mk >> pa) & 1) acc ^= xc[pa];
+                    } else {
+#pragma unroll
+                        for (int pa = 0; pa < 8; pa++) {
+                            const uint32_t mm = 0u - ((mk >> pa) & 1u);
+                            acc = (uint32_t)__builtin_amdgcn_bitop3_b32(
+                                xc[pa], mm, acc, 0x6a); /* (x & m) ^ acc */
+                        }
                     }
                     accp[i][pb] = acc;
                 }
@@ -1652,8 +1661,12 @@ hipError_t mec_launch_gf_matmul(const GfMatmulArgs *args, int n_dst, int n,
         dim3 blk(256);
 #define CASEB(E)                                                             \
     case E:                                                                  \
-        hipLaunchKernelGGL((gf_matmul_bs_kernel<E>), grid, blk, 0, stream,   \
-                           *args);                                           \
+        if (env_bs >= 2)                                                     \
+            hipLaunchKernelGGL((gf_matmul_bs_kernel<E, 1>), grid, blk, 0,    \
+                               stream, *args);                               \
+        else                                                                 \
+            hipLaunchKernelGGL((gf_matmul_bs_kernel<E, 0>), grid, blk, 0,    \
+                               stream, *args);                               \
         return hipGetLastError();
         switch (n_dst) {
             CASEB(1) CASEB(2) CASEB(3) CASEB(4) CASEB(5) CASEB(6) CASEB(7)

@@ -1,3 +1,6 @@
+# Debug probe (r2): reproduces and localizes fused3 parity mismatches —
+# prints (block, parity-row, first/last bad offset, count) per trial.
+# Used to find the launcher/kernel G mismatch.
 import os, sys
 sys.path.insert(0, os.environ.get("R", "/root/repo"))
 import minio_amd, oracle
