@@ -724,8 +724,8 @@ __device__ __forceinline__ void hh1_update(HH1 &s, uint64_t w, uint32_t S3) {
     }
 }
 
-template <bool RAGGED>
-__global__ void __launch_bounds__(256) hh256_batch4_kernel(HashArgs a) {
+template <bool RAGGED, int WG = 256>
+__global__ void __launch_bounds__(WG) hh256_batch4_kernel(HashArgs a) {
     const int64_t tid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
     const int64_t chain = tid >> 2;
     const int j = (int)(tid & 3); /* this lane's HighwayHash lane index */
@@ -1813,15 +1813,30 @@ hipError_t mec_launch_hash(int algo, const HashArgs *args,
             static const int hh4 = gf_env_int("MEC_HH4", 1);
             if (hh4 && !use_lds) {
                 /* 4-lane-per-chain kernel (see hh256_batch4_kernel):
-                 * 2x the waves of the pair kernel -> ~2x engaged CUs */
-                dim3 hblk(256);
-                grid.x = (uint32_t)((args->n_chains * 4 + 255) / 256);
-                if (args->msg_len % 32 == 0)
-                    hipLaunchKernelGGL((hh256_batch4_kernel<false>), grid,
-                                       hblk, 0, stream, *args);
-                else
-                    hipLaunchKernelGGL((hh256_batch4_kernel<true>), grid,
-                                       hblk, 0, stream, *args);
+                 * 2x the waves of the pair kernel -> ~2x engaged CUs.
+                 * MEC_HH4_WG: workgroup size — at the headline chain
+                 * count a 256-thread WG yields 192 WGs = 192 CUs (64
+                 * idle); 64-thread WGs spread the same waves over all
+                 * 256 CUs (r2 sweep). */
+                static const int wg4 = gf_env_int("MEC_HH4_WG", 256);
+                const int64_t lanes = args->n_chains * 4;
+#define HH4L(WGV)                                                            \
+                {                                                            \
+                    dim3 hblk(WGV);                                          \
+                    grid.x = (uint32_t)((lanes + WGV - 1) / WGV);            \
+                    if (args->msg_len % 32 == 0)                             \
+                        hipLaunchKernelGGL(                                  \
+                            (hh256_batch4_kernel<false, WGV>), grid, hblk,   \
+                            0, stream, *args);                               \
+                    else                                                     \
+                        hipLaunchKernelGGL(                                  \
+                            (hh256_batch4_kernel<true, WGV>), grid, hblk,    \
+                            0, stream, *args);                               \
+                }
+                if (wg4 <= 64) HH4L(64)
+                else if (wg4 <= 128) HH4L(128)
+                else HH4L(256)
+#undef HH4L
                 break;
             }
             if (use_lds && args->msg_len >= 512) {
