@@ -500,7 +500,10 @@ def test_stream_boundary_cases():
     # not 16-B aligned — the device row buffer placed after it must round
     # up to keep the row kernels' uint4 alignment invariant (regression:
     # scatter/interleave used misaligned uint4 at d=4,p=2,S=10)
-    for (d2, p2, bs2) in ((4, 2, 40), (3, 2, 51)):
+    # (3,2,144): S=48 -> stride 64; the interleave's 16-B gather window
+    # would read past the row stride on the final units without its
+    # base+32<=stride guard
+    for (d2, p2, bs2) in ((4, 2, 40), (3, 2, 51), (3, 2, 144)):
         with minio_amd.Erasure(d2, p2, bs2) as e2:
             for nfull in (1, 2, 5):
                 total_len = nfull * bs2 + 7
