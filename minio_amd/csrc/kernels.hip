@@ -2,23 +2,28 @@
  *
  * Replaces the compute of:
  *  - reedsolomon.Encoder.Encode / Reconstruct (GF(2^8) shard arithmetic;
- *    reference call sites cmd/erasure-coding.go:85,106,112) — gf_matmul
- *    kernels: byte-wise GF(2^8) constant-multiply as an in-register SWAR
- *    xtime ladder (poly 0x11D) over 16-byte lanes, wave-uniform coefficient
- *    bits steering scalar branches.  No MFMA: this is byte/integer work,
- *    HBM-bound by design.
+ *    reference call sites cmd/erasure-coding.go:85,106,112) — r2 default:
+ *    BIT-SLICED kernels (gf_encode_bs_kernel constexpr networks,
+ *    gf_matmul_bs_kernel runtime bit-matrices; gf_bs.h); the r1 SWAR
+ *    xtime-ladder kernels remain behind MEC_GF_BS=0 / MEC_GFM_BS=0.
+ *    No MFMA: this is byte/integer work.
  *  - streamingBitrotWriter's per-shard hash (cmd/bitrot-streaming.go:57-59)
- *    and BitrotAlgorithm.New digests (cmd/bitrot.go:47-64) — one hash chain
- *    per lane (HighwayHash-256 with the magic key, SHA-256, BLAKE2b-512);
- *    hashing is sequential per shard, parallelism comes from shards x
- *    blocks (SURVEY.md §7 hard part (b)).
+ *    and BitrotAlgorithm.New digests (cmd/bitrot.go:47-64) — HighwayHash
+ *    at 4 GPU lanes per chain (hh256_batch4_kernel, r2 default; r1
+ *    pair-lane kernel behind MEC_HH4=0), SHA-256 (16-round body),
+ *    BLAKE2b-512; hashing is sequential per shard, parallelism comes
+ *    from shards x blocks x lanes-per-chain (SURVEY.md §7 hard part (b)).
  *
- * Shipped structure: encode and hash run as two kernels, cross-batch
- * pipelined (batch t's hash overlaps batch t+1's GF on a second stream) —
- * measured at the joint memory wall (~4.4 TB/s).  Single-
- * pass fused variants (fused.hip lockstep, fused2.hip producer/consumer
- * LDS ring) are in-tree, bit-exact, and measured slower; the C-ABI
- * signature treats the pair as one fused operation either way.
+ * Shipped structure (r2): encode and hash run as two kernels back to
+ * back — the bit-sliced GF encode at ~90% of the HBM floor and the
+ * 4-lane HighwayHash at ~97% of its memory floor while simultaneously at
+ * the per-wave VALU issue cadence (DESIGN.md §4,
+ * profiles/probes_r2_valu.txt).  r1's cross-batch pipelining is kept
+ * behind MEC_PIPE=1 (overlap stopped paying once both legs reached their
+ * resource floors), and the single-pass fused variants (fused.hip
+ * lockstep, fused2.hip LDS ring, fused3.hip L2 handoff) are in-tree,
+ * bit-exact, measured, and opt-in; the C-ABI signature treats the pair
+ * as one fused operation either way.
  */
 #include <hip/hip_runtime.h>
 #include <cstdint>
