@@ -237,7 +237,8 @@ __global__ void __launch_bounds__(WAVES * 64) fused3_encode_hh_kernel(
                     __builtin_nontemporal_store(vhi, (v4u *)(orow + 16));
                 }
             }
-            if ((it + 1) % PUBK == 0 || it + 1 == n_iter) {
+            if (a.probe != 4 &&
+                ((it + 1) % PUBK == 0 || it + 1 == n_iter)) {
                 __builtin_amdgcn_s_waitcnt(0x0f70); /* vmcnt(0) */
                 if ((tid & 63) == 0)
                     __hip_atomic_store(&prog[wid], (int)it + 1,
