@@ -161,3 +161,24 @@ def test_simd_reconstruct_matches_scalar():
             assert rc == 0
             for i in range(d + p):
                 assert bufs[i].raw[:S] == want[i], (d, p, erased, i)
+
+
+def test_encode_linearity_full_size():
+    """GF(2^8) encode is linear over XOR: encode(x^y) == encode(x) ^
+    encode(y), checked at the FULL headline shard size (131072 B) — a
+    size-independent property that pins full-size behavior beyond the
+    small golden vectors (contract: properties at BASELINE sizes)."""
+    d, p = 8, 4
+    bs = 1 << 20
+    rs = oracle.RS(d, p)
+    x = oracle.fill_random(bs, 0xAAAA)
+    y = oracle.fill_random(bs, 0xBBBB)
+    xy = bytes(a ^ b for a, b in zip(x, y))
+    ex, ey, exy = (rs.encode_data(v) for v in (x, y, xy))
+    for s in range(d + p):
+        want = bytes(a ^ b for a, b in zip(ex[s], ey[s]))
+        assert exy[s] == want, s
+    # systematic prefix: data shards are the split input verbatim
+    S = len(ex[0])
+    for k in range(d):
+        assert ex[k] == x[k * S:(k + 1) * S].ljust(S, b"\0"), k
