@@ -642,7 +642,8 @@ def test_stream_pipeline_fuzz_geometries():
     byte-compared against the oracle or the original data."""
     rng = random.Random(0xF022)
     cases = []
-    for _ in range(10):
+    n_cases = int(os.environ.get("MEC_FUZZ_CASES", "10"))
+    for _ in range(n_cases):
         d = rng.randint(2, 16)
         p = rng.randint(1, min(8, d))
         bs = rng.choice([d * rng.randint(100, 5000),
