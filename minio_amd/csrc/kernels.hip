@@ -1430,18 +1430,30 @@ __global__ void __launch_bounds__(256) blake2b512_batch_kernel(HashArgs a) {
     for (int i = 0; i < 8; i++) h[i] = B2B_IV[i];
     h[0] ^= 0x01010000ull ^ 64;
     int64_t off = 0;
-    uint64_t m[16];
+    /* message double-buffer (r2, same treatment as SHA r16): the next
+     * 128-B block's loads ride under this block's 12 rounds */
+    uint64_t mA[16], mB[16];
+#define B2B_LOAD(M, OFF)                                                     \
+    {                                                                        \
+        const uint4 *p = (const uint4 *)(msg + (OFF));                       \
+        _Pragma("unroll") for (int q = 0; q < 8; q++) {                      \
+            uint4 v = p[q];                                                  \
+            M[2 * q] = (uint64_t)v.x | ((uint64_t)v.y << 32);                \
+            M[2 * q + 1] = (uint64_t)v.z | ((uint64_t)v.w << 32);            \
+        }                                                                    \
+    }
+    if (a.msg_len - off > 128) B2B_LOAD(mA, off)
     while (a.msg_len - off > 128) {
-        const uint4 *p = (const uint4 *)(msg + off);
+        if (a.msg_len - (off + 128) > 128) B2B_LOAD(mB, off + 128)
+        b2b_compress(h, mA, (uint64_t)(off + 128), false);
+        if (a.msg_len - (off + 128) > 128) {
 #pragma unroll
-        for (int q = 0; q < 8; q++) {
-            uint4 v = p[q];
-            m[2 * q] = (uint64_t)v.x | ((uint64_t)v.y << 32);
-            m[2 * q + 1] = (uint64_t)v.z | ((uint64_t)v.w << 32);
+            for (int i = 0; i < 16; i++) mA[i] = mB[i];
         }
-        b2b_compress(h, m, (uint64_t)(off + 128), false);
         off += 128;
     }
+#undef B2B_LOAD
+    uint64_t m[16];
     {
         const int rem = (int)(a.msg_len - off);
         uint8_t tail[128];
