@@ -26,6 +26,11 @@
  * tails (len % 32 != 0) are pinned transitively: the golden vectors cover
  * only multiples of 32; ragged behaviour follows the published portable
  * reference and is cross-checked oracle-vs-HIP in tests.
+ * r2 hardening (external ragged vectors remain unobtainable offline —
+ * DESIGN.md §2): the GPU cross-check covers EVERY len%32 residue, and a
+ * second independently-written AVX2 main loop (simd.c) agrees with this
+ * restatement for lengths 0..63 (test_simd_matches_scalar); the two share
+ * only the UpdateRemainder code, which stays single-sourced.
  */
 #ifndef MINIO_AMD_ORACLE_H
 #define MINIO_AMD_ORACLE_H
