@@ -77,7 +77,14 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     dist = None
+    saved_stdout = None
     if world > 1:
+        # Gloo prints connection banners to STDOUT, which would pollute
+        # the contract's single JSON line — shield fd 1 until the final
+        # print (everything stray goes to stderr instead)
+        sys.stdout.flush()
+        saved_stdout = os.dup(1)
+        os.dup2(2, 1)
         import torch.distributed as tdist
         tdist.init_process_group(backend="gloo")
         dist = tdist
@@ -317,6 +324,10 @@ def main():
                           f"({'reconstruct' if is_decode else 'encode+bitrot'})",
             }
 
+    if saved_stdout is not None:
+        sys.stdout.flush()
+        os.dup2(saved_stdout, 1)
+        os.close(saved_stdout)
     if rank == 0:
         out = {
             "metric": "GiB/s erasure encode+bitrot (EC8+4, 1 MiB blocks) per GPU and whole node"
