@@ -399,10 +399,16 @@ __global__ void __launch_bounds__(256) gf_matmul_bs_kernel(GfMatmulArgs a) {
         for (int i = 0; i < E; i++) {
             bs_transpose(accp[i]);
             uint8_t *orow = obase + (int64_t)a.dst_rows[i] * a.row_stride + j;
-            *(uint4 *)orow = uint4{accp[i][0], accp[i][1], accp[i][2],
-                                   accp[i][3]};
-            *(uint4 *)(orow + 16) = uint4{accp[i][4], accp[i][5],
-                                          accp[i][6], accp[i][7]};
+            /* nontemporal: rebuilt rows are written once and never
+             * re-read by this kernel (reads and writes hit the SAME
+             * in-place buffer — keeping stores out of the cache reduces
+             * the read/write turnaround that parks 42% of wave-cycles,
+             * gpurun decsq SQ pass) */
+            typedef unsigned int v4u __attribute__((ext_vector_type(4)));
+            v4u vlo = {accp[i][0], accp[i][1], accp[i][2], accp[i][3]};
+            v4u vhi = {accp[i][4], accp[i][5], accp[i][6], accp[i][7]};
+            __builtin_nontemporal_store(vlo, (v4u *)orow);
+            __builtin_nontemporal_store(vhi, (v4u *)(orow + 16));
         }
     }
 }
